@@ -1,0 +1,181 @@
+#!/usr/bin/env python3
+"""Benchmark harness (driver contract).
+
+Measures the BASELINE.json headline: images/sec (whole job) of ResNet-50
+BYOL training at 224x224, per-GPU batch 512 (global 4096 on 8 GPUs), fp32,
+synthetic data, random-init weights.  ``--gpus N`` is weak scaling: per-GPU
+batch stays fixed.
+
+Launch (driver): N=1 -> `python bench.py --gpus 1 --steps K --warmup W`;
+N>1 -> `python -m torch.distributed.run --nnodes=1 --nproc-per-node N
+--master-addr 127.0.0.1 ... bench.py --gpus N ...` (one rank per GPU over
+RCCL; RANK/LOCAL_RANK/WORLD_SIZE read from the env).
+
+The timed step is the FULL training step: forward (4 encoder passes + probe
++ fused EMA update), BYOL loss + CE, backward with bucketed RCCL all-reduce
+overlap, fused LARS+momentum step.  Nothing is skipped or cached.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+
+
+def parse():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--batch-per-gpu", type=int, default=512)
+    p.add_argument("--arch", type=str, default="resnet50")
+    p.add_argument("--image-size", type=int, default=224)
+    p.add_argument("--classes", type=int, default=1000)
+    p.add_argument("--dtype", type=str, default="fp32",
+                   choices=["fp32", "bf16"])
+    p.add_argument("--sync-bn", action="store_true", default=False)
+    p.add_argument("--channels-last", action="store_true", default=False)
+    p.add_argument("--bucket-cap-mb", type=float, default=32.0)
+    return p.parse_args()
+
+
+def main():
+    args = parse()
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    assert torch.cuda.is_available(), "bench.py needs a GPU"
+    torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+    distributed = world > 1
+    if distributed:
+        dist.init_process_group(backend="nccl")
+
+    from byol_amd import layers
+    from byol_amd.models.byol import BYOL
+    from byol_amd.objective import loss_function
+    from byol_amd.optim.lars import LARS
+    from byol_amd.parallel.ddp import FlatDDP
+    from byol_amd.parallel.sync_bn import convert_sync_batchnorm
+
+    torch.manual_seed(1234 + rank)
+    torch.backends.cudnn.benchmark = True
+
+    repr_size = {"resnet18": 512, "resnet34": 512}.get(args.arch, 2048)
+    model = BYOL(arch=args.arch, base_network_output_size=repr_size,
+                 projection_output_size=256,
+                 classifier_output_size=args.classes,
+                 total_training_steps=100000)
+    if args.sync_bn and distributed:
+        model = convert_sync_batchnorm(model)
+    model = model.to(device)
+    if args.channels_last:
+        model.base_network.to(memory_format=torch.channels_last)
+    model.finalize()
+    model.train()
+
+    inner = torch.optim.SGD(layers.add_weight_decay(model, 1e-6),
+                            lr=0.2 * (args.batch_per_gpu * world / 256),
+                            momentum=0.9)
+    opt = LARS(inner, eps=0.0)
+    opt.attach_flat_space(model.flat_space)
+
+    net = model
+    if distributed:
+        net = FlatDDP(model, bucket_cap_mb=args.bucket_cap_mb)
+
+    B = args.batch_per_gpu
+    S = args.image_size
+    # synthetic two-view batches, pre-generated on device (data=synthetic;
+    # augmentation happens upstream of the timed step in real training too
+    # when the GPU pipeline is used)
+    n_pool = 2
+    gen = torch.Generator(device="cuda").manual_seed(99 + rank)
+    pool = [(torch.rand(B, 3, S, S, device=device, generator=gen),
+             torch.rand(B, 3, S, S, device=device, generator=gen),
+             torch.randint(args.classes, (B,), device=device, generator=gen))
+            for _ in range(n_pool)]
+    if args.channels_last:
+        pool = [(a.to(memory_format=torch.channels_last),
+                 b.to(memory_format=torch.channels_last), l)
+                for a, b, l in pool]
+
+    use_bf16 = args.dtype == "bf16"
+
+    def step(i):
+        a1, a2, lab = pool[i % n_pool]
+        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
+            out = net(a1, a2)
+            loss = loss_function(
+                out["online_prediction1"].float(),
+                out["online_prediction2"].float(),
+                out["target_projection1"].float(),
+                out["target_projection2"].float())
+            loss = loss + F.cross_entropy(out["linear_preds"].float(),
+                                          torch.cat([lab, lab]))
+        opt.zero_grad()
+        loss.backward()
+        if distributed:
+            net.finish_grad_sync()
+        opt.step()
+        return loss
+
+    for i in range(args.warmup):
+        step(i)
+
+    if distributed:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    torch.cuda.synchronize()
+    if distributed:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    t = torch.tensor([elapsed], device=device)
+    if distributed:
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t)
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    # whole-job throughput: each step consumes B images per GPU (two
+    # augmented views of each)
+    images_per_sec = args.batch_per_gpu * world * args.steps / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec (whole node) ResNet-50 BYOL bs=4096 "
+                      "224px at 1/2/4/8 MI355X",
+            "value": images_per_sec,
+            "unit": "images/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.arch,
+                "global_batch": args.batch_per_gpu * world,
+                "image_size": args.image_size,
+                "parallelism": f"dp{world}",
+                "sync_bn": args.sync_bn,
+                "channels_last": args.channels_last,
+            },
+        }))
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,336 @@
+"""Training engine: setup orchestration + the shared train/eval minibatch
+loop, replicating the reference's ``execute_graph``/``train``/``test``/``run``
+semantics (``/root/reference/main.py:403-499,559-692,732-783``) on the
+MI355X-native stack (flat-param BYOL, FlatDDP over RCCL, custom SyncBN,
+fused loss/EMA/LARS HIP kernels)."""
+
+import functools
+import pprint
+import time
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .. import layers
+from ..data import build_train_and_test_transforms, get_loader
+from ..models.byol import BYOL
+from ..objective import loss_function
+from ..optim import build_optimizer
+from ..parallel import FlatDDP, convert_sync_batchnorm
+from . import metrics
+from .grapher import Grapher
+from .saver import CheckpointBundle, ModelSaver, get_name
+
+__all__ = ["build_loader_model_grapher", "execute_graph", "train", "test",
+           "run", "register_plots", "register_images", "make_grid"]
+
+
+def _dummy_context():
+    import contextlib
+    return contextlib.nullcontext()
+
+
+# ---------------------------------------------------------------------------
+# setup
+# ---------------------------------------------------------------------------
+
+def build_loader_model_grapher(args):
+    """transforms -> loader -> derived sizes -> BYOL -> (SyncBN) -> device ->
+    lazy sanity pass -> weight init -> finalize flat space -> DDP -> grapher
+    (order mirrors ``/root/reference/main.py:403-462``; finalize happens after
+    all structural changes so the flat buffer is terminal)."""
+    train_transform, test_transform = build_train_and_test_transforms(args)
+    loader = get_loader(train_transform=train_transform,
+                        test_transform=test_transform, **vars(args))
+
+    args.input_shape = loader.input_shape
+    args.num_train_samples = loader.num_train_samples // args.num_replicas
+    args.num_test_samples = loader.num_test_samples  # test is not sharded
+    args.num_valid_samples = loader.num_valid_samples // args.num_replicas
+    args.steps_per_train_epoch = args.num_train_samples // args.batch_size
+    args.total_train_steps = args.epochs * max(args.steps_per_train_epoch, 1)
+
+    network = BYOL(arch=args.arch,
+                   base_network_output_size=args.representation_size,
+                   projection_output_size=args.projection_size,
+                   classifier_output_size=loader.output_size,
+                   total_training_steps=args.total_train_steps,
+                   head_latent_size=args.head_latent_size,
+                   base_decay=args.base_decay,
+                   in_channels=loader.input_shape[0])
+    if args.convert_to_sync_bn and args.num_replicas > 1:
+        network = convert_sync_batchnorm(network)
+    if args.cuda:
+        network = network.cuda()
+    network = layers.init_weights(network, init=args.weight_initialization)
+    network.finalize()  # flat param space + EMA prime (terminal)
+    lazy_generate_modules(network, loader.train_loader, args)
+
+    if args.num_replicas > 1:
+        network = FlatDDP(network,
+                          bucket_cap_mb=getattr(args, "bucket_cap_mb", 32.0),
+                          broadcast_buffers=getattr(
+                              args, "broadcast_buffers", False))
+
+    print("model has {:.3f} million parameters.".format(
+        layers.number_of_parameters(network) / 1e6))
+
+    grapher = None
+    if args.distributed_rank == 0:
+        if args.visdom_url is not None:
+            grapher = Grapher("visdom", env=get_name(args),
+                              server=args.visdom_url, port=args.visdom_port,
+                              log_folder=args.log_dir)
+        else:
+            import os
+            grapher = Grapher("tensorboard",
+                              logdir=os.path.join(args.log_dir,
+                                                  get_name(args)))
+    return loader, network, grapher
+
+
+def lazy_generate_modules(model, loader, args):
+    """One eval-mode sanity forward: prints shapes/dtypes and hard-errors if
+    augmented pixels leave [0, 1] (``/root/reference/main.py:465-499``)."""
+    model.eval()
+    for augmentation1, augmentation2, labels in loader:
+        with torch.no_grad():
+            print("augmentation1 = {} / {} | augmentation2 = {} / {} | "
+                  "labels = {} / {}".format(
+                      tuple(augmentation1.shape), augmentation1.dtype,
+                      tuple(augmentation2.shape), augmentation2.dtype,
+                      tuple(labels.shape), labels.dtype))
+            a1_min, a1_max = augmentation1.min(), augmentation1.max()
+            a2_min, a2_max = augmentation2.min(), augmentation2.max()
+            print(f"aug1 in range [min: {a1_min}, max: {a1_max}] | "
+                  f"aug2 in range [min: {a2_min}, max: {a2_max}]")
+            if a1_max > 1.0 or a1_min < 0:
+                raise ValueError("aug1 outside [0, 1]. You probably dont "
+                                 "want this.")
+            if a2_max > 1.0 or a2_min < 0:
+                raise ValueError("aug2 outside [0, 1]. You probably dont "
+                                 "want this.")
+            if args.cuda:
+                augmentation1 = augmentation1.cuda(non_blocking=True)
+                augmentation2 = augmentation2.cuda(non_blocking=True)
+            _ = model(augmentation1, augmentation2)
+            break
+    if args.polyak_ema > 0:
+        layers.polyak_ema_parameters(model, args.polyak_ema)
+
+
+# ---------------------------------------------------------------------------
+# plot/image registration (key-name conventions of the reference)
+# ---------------------------------------------------------------------------
+
+def make_grid(images: torch.Tensor, nrow: int = 8,
+              normalize: bool = True) -> torch.Tensor:
+    """Minimal image-grid builder (torchvision.utils.make_grid replacement)."""
+    imgs = images.detach().float().cpu()
+    if normalize:
+        lo, hi = imgs.amin(dim=(1, 2, 3), keepdim=True), \
+            imgs.amax(dim=(1, 2, 3), keepdim=True)
+        imgs = (imgs - lo) / (hi - lo).clamp(min=1e-8)
+    n, c, h, w = imgs.shape
+    ncol = min(nrow, n)
+    nrows = (n + ncol - 1) // ncol
+    grid = torch.zeros(c, nrows * (h + 2) + 2, ncol * (w + 2) + 2)
+    for idx in range(n):
+        r, col = divmod(idx, ncol)
+        grid[:, 2 + r * (h + 2):2 + r * (h + 2) + h,
+             2 + col * (w + 2):2 + col * (w + 2) + w] = imgs[idx]
+    return grid
+
+
+def register_plots(loss, grapher, epoch, args, prefix="train"):
+    if args.distributed_rank != 0 or grapher is None:
+        return
+    import numpy as np
+    for k, v in loss.items():
+        if isinstance(v, dict):
+            register_plots(v, grapher, epoch, args, prefix=prefix)
+        if "mean" in k or "scalar" in k:
+            key_name = "-".join(k.split("_")[0:-1])
+            value = v.item() if torch.is_tensor(v) else float(v)
+            grapher.add_scalar(f"{prefix}_{key_name}", value, epoch)
+
+
+def register_images(output_map, grapher, args, prefix="train"):
+    if args.distributed_rank != 0 or grapher is None:
+        return
+    for k, v in output_map.items():
+        if isinstance(v, dict):
+            register_images(v, grapher, args, prefix=prefix)
+        if "img" in k or "imgs" in k:
+            key_name = "-".join(k.split("_")[0:-1])
+            grapher.add_image(f"{prefix}_{key_name}",
+                              make_grid(v, normalize=True), global_step=0)
+
+
+# ---------------------------------------------------------------------------
+# the engine loop
+# ---------------------------------------------------------------------------
+
+def _sum_scalars(d1: dict, d2: dict) -> dict:
+    out = {}
+    for k, v in d2.items():
+        prev = d1.get(k, 0.0)
+        vd = v.detach() if torch.is_tensor(v) else v
+        out[k] = prev + vd
+    return out
+
+
+def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
+                  prefix="test"):
+    start_time = time.time()
+    is_eval = "train" not in prefix
+    model.eval() if is_eval else model.train()
+    assert optimizer is None if is_eval else optimizer is not None
+    loss_map, num_samples = {}, 0
+    num_minibatches = 0
+    is_ddp = isinstance(model, FlatDDP)
+    autocast_ctx = (torch.autocast("cuda", dtype=torch.bfloat16)
+                    if args.half and args.cuda else _dummy_context())
+
+    for num_minibatches, (augmentation1, augmentation2, labels) in \
+            enumerate(loader):
+        if args.cuda:
+            augmentation1 = augmentation1.cuda(non_blocking=True)
+            augmentation2 = augmentation2.cuda(non_blocking=True)
+            labels = labels.cuda(non_blocking=True)
+
+        with torch.no_grad() if is_eval else _dummy_context():
+            with autocast_ctx:
+                if is_eval and args.polyak_ema > 0:
+                    output_dict = layers.get_polyak_prediction(
+                        model, pred_fn=functools.partial(
+                            model, augmentation1, augmentation2))
+                else:
+                    output_dict = model(augmentation1, augmentation2)
+
+                byol_loss = loss_function(
+                    online_prediction1=output_dict["online_prediction1"].float(),
+                    online_prediction2=output_dict["online_prediction2"].float(),
+                    target_projection1=output_dict["target_projection1"].float(),
+                    target_projection2=output_dict["target_projection2"].float())
+                classifier_labels = labels if is_eval else \
+                    torch.cat([labels, labels], 0)
+                classifier_loss = F.cross_entropy(
+                    input=output_dict["linear_preds"].float(),
+                    target=classifier_labels)
+            acc1, acc5 = metrics.topk(output=output_dict["linear_preds"],
+                                      target=classifier_labels, topk=(1, 5))
+            loss_t = {
+                "loss_mean": byol_loss + classifier_loss,
+                "byol_loss_mean": byol_loss,
+                "linear_loss_mean": classifier_loss,
+                "top1_mean": acc1,
+                "top5_mean": acc5,
+            }
+            detached_t = {k: (v.detach() if torch.is_tensor(v) else v)
+                          for k, v in loss_t.items()}
+            loss_map = detached_t if not loss_map else \
+                _sum_scalars(loss_map, detached_t)
+            num_samples += augmentation1.size(0)
+
+        if not is_eval:
+            optimizer.zero_grad()
+            loss_t["loss_mean"].backward()
+            if is_ddp:
+                model.finish_grad_sync()
+            if args.clip > 0:
+                nn.utils.clip_grad_value_(model.parameters(), args.clip)
+            optimizer.step()
+            if args.polyak_ema > 0:
+                layers.polyak_ema_parameters(model, args.polyak_ema)
+            del loss_t
+
+        if args.debug_step:
+            break
+
+    loss_map = {k: v / (num_minibatches + 1) for k, v in loss_map.items()}
+
+    to_log = ("{}-{}[Epoch {}][{} samples][{:.2f} sec]:\tLoss: {:.4f}\t"
+              "Top-1: {:.4f}\tTop-5: {:.4f}")
+    print(to_log.format(
+        prefix, args.distributed_rank, epoch, num_samples,
+        time.time() - start_time,
+        float(loss_map["loss_mean"]),
+        float(loss_map["top1_mean"]),
+        float(loss_map["top5_mean"])))
+
+    register_plots(dict(loss_map), grapher, epoch=epoch, args=args,
+                   prefix=prefix)
+
+    num_images_to_post = min(64, augmentation1.shape[0])
+    image_size_to_post = min(64, augmentation1.shape[-1])
+    image_map = {
+        "augmentation1_imgs": F.interpolate(
+            augmentation1[0:num_images_to_post].float(),
+            size=(image_size_to_post, image_size_to_post)),
+        "augmentation2_imgs": F.interpolate(
+            augmentation2[0:num_images_to_post].float(),
+            size=(image_size_to_post, image_size_to_post)),
+    }
+    register_images(image_map, grapher, args, prefix=prefix)
+    if grapher is not None:
+        grapher.save()
+
+    loss_val = float(loss_map["loss_mean"])
+    loss_map.clear()
+    return loss_val
+
+
+def train(epoch, model, optimizer, train_loader, grapher, args):
+    return execute_graph(epoch, model, train_loader, grapher, args,
+                         optimizer, prefix="train")
+
+
+def test(epoch, model, test_loader, grapher, args):
+    return execute_graph(epoch, model, test_loader, grapher, args,
+                         prefix="test")
+
+
+# ---------------------------------------------------------------------------
+# epoch driver
+# ---------------------------------------------------------------------------
+
+def run(rank, args):
+    from ..parallel import init_multiprocessing_and_cuda
+    init_multiprocessing_and_cuda(rank, args)
+    loader, model, grapher = build_loader_model_grapher(args)
+    print(pprint.PrettyPrinter(indent=4).pformat(vars(args)))
+    optimizer, scheduler = build_optimizer(model, args)
+
+    bundle = CheckpointBundle(model, optimizer, scheduler, args)
+    saver = ModelSaver(bundle, early_stop=args.early_stop,
+                       rank=args.distributed_rank,
+                       burn_in_interval=int(0.1 * args.epochs),
+                       larger_is_better=False, max_early_stop_steps=10)
+    restore_dict = saver.restore()
+    init_epoch = restore_dict["epoch"]
+
+    for epoch in range(init_epoch, args.epochs + 1):
+        train(epoch, model, optimizer, loader.train_loader, grapher, args)
+        test_loss = test(epoch, model, loader.test_loader, grapher, args)
+        loader.set_all_epochs(epoch)
+
+        scheduler.step()
+        register_plots(
+            {"learning_rate_scalar": optimizer.param_groups[0]["lr"]},
+            grapher, epoch, args)
+
+        if saver(test_loss):
+            saver.restore()
+            test_loss = test(epoch, model, loader.test_loader, grapher, args)
+            break
+
+        if epoch == 2 and args.distributed_rank == 0 and grapher is not None:
+            grapher.add_text(
+                "config",
+                pprint.PrettyPrinter(indent=4).pformat(vars(args)), 0)
+
+    if grapher is not None:
+        grapher.close()

@@ -1,0 +1,160 @@
+"""ResNet encoder family, written from scratch for the MI355X BYOL engine.
+
+Replicates the architecture the reference reaches through torchvision
+(``/root/reference/main.py:190-193`` builds ``nn.Sequential(*children()[:-1])``
+of a torchvision ResNet, i.e. the full network minus the final fc but keeping
+the global average pool).  We build exactly that encoder directly — there is no
+torchvision in this environment, and owning the module tree lets us fuse
+conv+BN+ReLU blocks into HIP kernels later without fighting a vendored model.
+
+Supported archs: resnet18/34/50/101/152/200 (the reference exposes every
+torchvision arch; the ResNet family is the one its README and defaults use).
+
+Defaults (stride/width/downsample placement, kaiming init, BN eps/momentum)
+match torchvision's ResNet so numerics line up with the reference oracle.
+"""
+
+from typing import List, Optional, Type, Union
+
+import torch
+import torch.nn as nn
+
+__all__ = ["ResNetEncoder", "build_encoder", "ARCH_SPECS", "arch_names"]
+
+
+def conv3x3(in_planes: int, out_planes: int, stride: int = 1) -> nn.Conv2d:
+    return nn.Conv2d(in_planes, out_planes, kernel_size=3, stride=stride,
+                     padding=1, bias=False)
+
+
+def conv1x1(in_planes: int, out_planes: int, stride: int = 1) -> nn.Conv2d:
+    return nn.Conv2d(in_planes, out_planes, kernel_size=1, stride=stride,
+                     bias=False)
+
+
+class BasicBlock(nn.Module):
+    expansion = 1
+
+    def __init__(self, inplanes: int, planes: int, stride: int = 1,
+                 downsample: Optional[nn.Module] = None):
+        super().__init__()
+        self.conv1 = conv3x3(inplanes, planes, stride)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.relu = nn.ReLU(inplace=True)
+        self.conv2 = conv3x3(planes, planes)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.downsample = downsample
+        self.stride = stride
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        identity = x
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.bn2(self.conv2(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        out = out + identity
+        return self.relu(out)
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, inplanes: int, planes: int, stride: int = 1,
+                 downsample: Optional[nn.Module] = None):
+        super().__init__()
+        self.conv1 = conv1x1(inplanes, planes)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = conv3x3(planes, planes, stride)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.conv3 = conv1x1(planes, planes * self.expansion)
+        self.bn3 = nn.BatchNorm2d(planes * self.expansion)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = downsample
+        self.stride = stride
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        identity = x
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        if self.downsample is not None:
+            identity = self.downsample(x)
+        out = out + identity
+        return self.relu(out)
+
+
+ARCH_SPECS = {
+    "resnet18": (BasicBlock, [2, 2, 2, 2]),
+    "resnet34": (BasicBlock, [3, 4, 6, 3]),
+    "resnet50": (Bottleneck, [3, 4, 6, 3]),
+    "resnet101": (Bottleneck, [3, 4, 23, 3]),
+    "resnet152": (Bottleneck, [3, 8, 36, 3]),
+    "resnet200": (Bottleneck, [3, 24, 36, 3]),
+}
+
+
+def arch_names() -> List[str]:
+    return sorted(ARCH_SPECS.keys())
+
+
+class ResNetEncoder(nn.Module):
+    """ResNet minus the final fc: stem -> 4 stages -> global avgpool.
+
+    Output is ``(B, out_channels, 1, 1)``; callers ``.view(-1, C)`` it, the
+    same contract as the reference's ``children()[:-1]`` Sequential
+    (``/root/reference/main.py:190-193, 238``).
+    """
+
+    def __init__(self, block: Type[Union[BasicBlock, Bottleneck]],
+                 layers: List[int], in_channels: int = 3):
+        super().__init__()
+        self.inplanes = 64
+        self.conv1 = nn.Conv2d(in_channels, 64, kernel_size=7, stride=2,
+                               padding=3, bias=False)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
+        self.layer1 = self._make_layer(block, 64, layers[0])
+        self.layer2 = self._make_layer(block, 128, layers[1], stride=2)
+        self.layer3 = self._make_layer(block, 256, layers[2], stride=2)
+        self.layer4 = self._make_layer(block, 512, layers[3], stride=2)
+        self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
+        self.out_channels = 512 * block.expansion
+
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.kaiming_normal_(m.weight, mode="fan_out",
+                                        nonlinearity="relu")
+            elif isinstance(m, nn.BatchNorm2d):
+                nn.init.constant_(m.weight, 1)
+                nn.init.constant_(m.bias, 0)
+
+    def _make_layer(self, block, planes: int, blocks: int,
+                    stride: int = 1) -> nn.Sequential:
+        downsample = None
+        if stride != 1 or self.inplanes != planes * block.expansion:
+            downsample = nn.Sequential(
+                conv1x1(self.inplanes, planes * block.expansion, stride),
+                nn.BatchNorm2d(planes * block.expansion),
+            )
+        layers = [block(self.inplanes, planes, stride, downsample)]
+        self.inplanes = planes * block.expansion
+        layers += [block(self.inplanes, planes) for _ in range(1, blocks)]
+        return nn.Sequential(*layers)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.relu(self.bn1(self.conv1(x)))
+        x = self.maxpool(x)
+        x = self.layer1(x)
+        x = self.layer2(x)
+        x = self.layer3(x)
+        x = self.layer4(x)
+        return self.avgpool(x)
+
+
+def build_encoder(arch: str, in_channels: int = 3) -> ResNetEncoder:
+    if arch not in ARCH_SPECS:
+        raise ValueError(
+            f"unknown arch {arch!r}; available: {', '.join(arch_names())}")
+    block, layers = ARCH_SPECS[arch]
+    return ResNetEncoder(block, layers, in_channels=in_channels)

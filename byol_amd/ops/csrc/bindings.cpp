@@ -1,0 +1,119 @@
+// Python bindings for the byol_amd CDNA4 HIP kernels (gfx950).
+#include <torch/extension.h>
+#include <cstdint>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+void launch_flat_ema_update(float* mean, const float* x, float decay,
+                            int64_t n, hipStream_t stream);
+void launch_byol_loss_forward(const float* p1, const float* p2,
+                              const float* z1, const float* z2, float* stats,
+                              float* loss, int64_t n, int64_t batch,
+                              hipStream_t stream);
+void launch_byol_loss_backward(const float* p1, const float* p2,
+                               const float* z1, const float* z2,
+                               const float* stats, const float* grad_out,
+                               float* g1, float* g2, int64_t n,
+                               int64_t batch, hipStream_t stream);
+void launch_lars_momentum_step(float* p, const float* g, float* m,
+                               float* norm_acc, float* alr,
+                               const int64_t* seg_off,
+                               const int64_t* seg_len, const float* seg_wd,
+                               const int* seg_adapt, const int* chunk_seg,
+                               const int64_t* chunk_base, int nseg,
+                               int nchunks, int64_t chunk, float trust,
+                               float eps, float lr, float momentum,
+                               int m_init, hipStream_t stream);
+
+namespace {
+
+#define CHECK_IN(t)                                                        \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                        \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous");              \
+  TORCH_CHECK((t).scalar_type() == at::kFloat, #t " must be fp32")
+
+void flat_ema_update(torch::Tensor mean, torch::Tensor x, double decay) {
+  CHECK_IN(mean);
+  CHECK_IN(x);
+  TORCH_CHECK(mean.numel() == x.numel(), "size mismatch");
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_flat_ema_update(mean.data_ptr<float>(), x.data_ptr<float>(),
+                         static_cast<float>(decay), mean.numel(), stream);
+}
+
+std::tuple<torch::Tensor, torch::Tensor> byol_loss_forward(
+    torch::Tensor p1, torch::Tensor p2, torch::Tensor z1, torch::Tensor z2) {
+  CHECK_IN(p1); CHECK_IN(p2); CHECK_IN(z1); CHECK_IN(z2);
+  TORCH_CHECK(p1.dim() == 2, "expected 2-D [B, D]");
+  TORCH_CHECK(p1.sizes() == p2.sizes() && p1.sizes() == z1.sizes() &&
+              p1.sizes() == z2.sizes(), "shape mismatch");
+  const int64_t n = p1.numel();
+  const int64_t batch = p1.size(0);
+  auto stats = torch::zeros({6}, p1.options());
+  auto loss = torch::empty({}, p1.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_byol_loss_forward(p1.data_ptr<float>(), p2.data_ptr<float>(),
+                           z1.data_ptr<float>(), z2.data_ptr<float>(),
+                           stats.data_ptr<float>(), loss.data_ptr<float>(), n,
+                           batch, stream);
+  return {loss, stats};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> byol_loss_backward(
+    torch::Tensor p1, torch::Tensor p2, torch::Tensor z1, torch::Tensor z2,
+    torch::Tensor stats, torch::Tensor grad_out) {
+  CHECK_IN(p1); CHECK_IN(p2); CHECK_IN(z1); CHECK_IN(z2); CHECK_IN(stats);
+  TORCH_CHECK(grad_out.is_cuda() && grad_out.scalar_type() == at::kFloat,
+              "grad_out must be fp32 on GPU");
+  const int64_t n = p1.numel();
+  const int64_t batch = p1.size(0);
+  auto g1 = torch::empty_like(p1);
+  auto g2 = torch::empty_like(p2);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_byol_loss_backward(
+      p1.data_ptr<float>(), p2.data_ptr<float>(), z1.data_ptr<float>(),
+      z2.data_ptr<float>(), stats.data_ptr<float>(),
+      grad_out.contiguous().data_ptr<float>(), g1.data_ptr<float>(),
+      g2.data_ptr<float>(), n, batch, stream);
+  return {g1, g2};
+}
+
+void lars_momentum_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                        torch::Tensor norm_acc, torch::Tensor alr,
+                        torch::Tensor seg_off, torch::Tensor seg_len,
+                        torch::Tensor seg_wd, torch::Tensor seg_adapt,
+                        torch::Tensor chunk_seg, torch::Tensor chunk_base,
+                        double trust, double eps, double lr, double momentum,
+                        int64_t m_init) {
+  CHECK_IN(p); CHECK_IN(g); CHECK_IN(m); CHECK_IN(norm_acc); CHECK_IN(alr);
+  TORCH_CHECK(seg_off.scalar_type() == at::kLong &&
+              seg_len.scalar_type() == at::kLong &&
+              chunk_base.scalar_type() == at::kLong, "bad index dtypes");
+  TORCH_CHECK(seg_adapt.scalar_type() == at::kInt &&
+              chunk_seg.scalar_type() == at::kInt, "bad flag dtypes");
+  const int nseg = seg_off.numel();
+  const int nchunks = chunk_seg.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_lars_momentum_step(
+      p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+      norm_acc.data_ptr<float>(), alr.data_ptr<float>(),
+      seg_off.data_ptr<int64_t>(), seg_len.data_ptr<int64_t>(),
+      seg_wd.data_ptr<float>(), seg_adapt.data_ptr<int>(),
+      chunk_seg.data_ptr<int>(), chunk_base.data_ptr<int64_t>(), nseg,
+      nchunks, 65536, static_cast<float>(trust), static_cast<float>(eps),
+      static_cast<float>(lr), static_cast<float>(momentum),
+      static_cast<int>(m_init), stream);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("flat_ema_update", &flat_ema_update,
+          "fused flat-parameter EMA update (gfx950)");
+  mod.def("byol_loss_forward", &byol_loss_forward,
+          "fused BYOL loss forward (gfx950)");
+  mod.def("byol_loss_backward", &byol_loss_backward,
+          "fused BYOL loss backward (gfx950)");
+  mod.def("lars_momentum_step", &lars_momentum_step,
+          "fused multi-tensor LARS+momentum step (gfx950)");
+}

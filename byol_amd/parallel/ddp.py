@@ -1,0 +1,153 @@
+"""Flat-buffer data-parallel wrapper, MI355X-native replacement for
+``DistributedDataParallelPassthrough`` (reference call site
+``/root/reference/main.py:440-443``).
+
+Design:
+
+* gradients already live in ONE contiguous flat buffer (``FlatParamSpace``),
+  partitioned into fixed buckets in reverse parameter order (the order
+  backward produces them); per-parameter post-accumulate-grad hooks launch an
+  async RCCL all-reduce per bucket as soon as its last gradient lands, so
+  communication overlaps the rest of backward — bucket size is a knob tuned
+  for 7 x ~153 GB/s point-to-point xGMI links (default 32 MiB);
+* gradient averaging is pre-scale (1/world) + SUM all-reduce — works on both
+  RCCL ("nccl") and gloo;
+* there is NO per-step buffer broadcast.  torch-1.5 DDP broadcast every
+  buffer each forward — for this model that is ~148.6 MB/step of redundant
+  xGMI traffic, because the CosEMA mean is a deterministic function of the
+  (synchronised) parameters and BN running stats are replica-identical in
+  expectation; ``broadcast_buffers=True`` restores reference behaviour for
+  parity experiments;
+* parameters (the flat buffer) are broadcast once at construction, exactly
+  like DDP init;
+* attribute access falls through to the wrapped module (the reference's
+  "Passthrough" behaviour), so saver hooks and ``model.flat_space`` work
+  through the wrapper;
+* parameters that received no gradient this iteration are still covered: the
+  engine calls :meth:`finish_grad_sync` before ``optimizer.step`` and any
+  bucket whose hooks never completed is all-reduced then (its stale region is
+  zero — ``zero_grads`` zeroes the whole flat buffer), which is the
+  ``find_unused_parameters=True`` semantics without the per-iteration graph
+  walk.
+"""
+
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+__all__ = ["FlatDDP"]
+
+
+class _Bucket:
+    __slots__ = ("start", "end", "param_ids", "pending", "work", "launched")
+
+    def __init__(self, start: int, end: int):
+        self.start = start
+        self.end = end
+        self.param_ids: List[int] = []
+        self.pending = 0
+        self.work = None
+        self.launched = False
+
+
+class FlatDDP(nn.Module):
+    def __init__(self, module: nn.Module, bucket_cap_mb: float = 32.0,
+                 process_group=None, broadcast_buffers: bool = False):
+        super().__init__()
+        self.module = module
+        space = module.flat_space  # requires finalize()d BYOL
+        self._space = space
+        self.process_group = process_group
+        self.world_size = dist.get_world_size(process_group)
+        self.broadcast_buffers = broadcast_buffers
+        self._grad_scale = 1.0 / self.world_size
+
+        # one-time parameter sync (flat buffer = ONE broadcast)
+        with torch.no_grad():
+            dist.broadcast(space.flat_params, src=0, group=process_group)
+            # EMA mean is derived from the initial params; re-prime so every
+            # replica's mean matches rank 0 (it was primed pre-broadcast).
+            mean = getattr(module.target_network, "mean", None)
+            if mean is not None:
+                dist.broadcast(mean, src=0, group=process_group)
+
+        # bucket partition: reverse parameter order, split at cap
+        cap_elems = int(bucket_cap_mb * 1024 * 1024 /
+                        space.flat_params.element_size())
+        self.buckets: List[_Bucket] = []
+        cur: Optional[_Bucket] = None
+        params = list(module.parameters())
+        layout = space.layout  # same order as parameters()
+        self._param_bucket = {}
+        for pid in reversed(range(len(layout))):
+            name, off, n, shape = layout[pid]
+            if cur is None or (cur.end - cur.start) + n > cap_elems:
+                cur = _Bucket(off, off + n)
+                self.buckets.append(cur)
+            else:
+                cur.start = off  # reverse order: extend downwards
+            cur.param_ids.append(pid)
+            self._param_bucket[pid] = cur
+
+        self._hook_handles = []
+        for pid, p in enumerate(params):
+            self._hook_handles.append(p.register_post_accumulate_grad_hook(
+                self._make_hook(pid)))
+        self._reset_buckets()
+
+    # -- hooks ------------------------------------------------------------
+    def _reset_buckets(self):
+        for b in self.buckets:
+            b.pending = len(b.param_ids)
+            b.work = None
+            b.launched = False
+
+    def _make_hook(self, pid: int):
+        def hook(_param):
+            if not self.require_backward_grad_sync:
+                return
+            b = self._param_bucket[pid]
+            b.pending -= 1
+            if b.pending == 0 and not b.launched:
+                self._launch(b)
+        return hook
+
+    def _launch(self, b: _Bucket):
+        view = self._space.flat_grads[b.start:b.end]
+        if self._grad_scale != 1.0:
+            view.mul_(self._grad_scale)
+        b.work = dist.all_reduce(view, op=dist.ReduceOp.SUM,
+                                 group=self.process_group, async_op=True)
+        b.launched = True
+
+    # -- public API -------------------------------------------------------
+    require_backward_grad_sync = True
+
+    def finish_grad_sync(self):
+        """Launch any not-yet-fired buckets and wait for all; call after
+        backward, before the optimizer step."""
+        for b in self.buckets:
+            if not b.launched:
+                self._launch(b)
+        for b in self.buckets:
+            if b.work is not None:
+                b.work.wait()
+        self._reset_buckets()
+
+    def forward(self, *args, **kwargs):
+        if self.broadcast_buffers and self.module.training:
+            with torch.no_grad():
+                for buf in self.module.buffers():
+                    if buf is not None and buf.numel():
+                        dist.broadcast(buf, src=0, group=self.process_group)
+        return self.module(*args, **kwargs)
+
+    # Passthrough: the reference's DistributedDataParallelPassthrough
+    # forwards attribute access to .module.
+    def __getattr__(self, name):
+        try:
+            return super().__getattr__(name)
+        except AttributeError:
+            return getattr(self.module, name)

@@ -1,0 +1,155 @@
+"""Cross-replica synchronized BatchNorm (our own, not torch's).
+
+The reference opts in via ``nn.SyncBatchNorm.convert_sync_batchnorm``
+(``/root/reference/main.py:433``, flag ``--convert-to-sync-bn``).  This
+implementation keeps those semantics — global-batch statistics in training,
+per-layer stat all-reduce in forward AND backward — but is built for xGMI:
+the per-layer stats travel as ONE packed ``2C+1`` float vector (sum, sumsq,
+count) instead of separate tensors, and the normalize/backward element work
+is shaped for a later fused HIP kernel swap-in.
+
+Works on any torch.distributed backend (RCCL on GPU, gloo on CPU — which is
+how the multi-process unit tests validate it against a single-process
+full-batch BatchNorm oracle; torch's own SyncBatchNorm is CUDA-only).
+"""
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+__all__ = ["SyncBatchNorm", "convert_sync_batchnorm"]
+
+
+def _flatten_to_nc(x: torch.Tensor):
+    """view (N, C, *) as dims for per-channel reduction"""
+    return [0] + list(range(2, x.dim()))
+
+
+class _SyncBNFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps, process_group, world_size):
+        reduce_dims = _flatten_to_nc(x)
+        count_local = x.numel() // x.shape[1]
+        sum_local = x.sum(dim=reduce_dims)
+        sqsum_local = (x * x).sum(dim=reduce_dims)
+        if world_size > 1:
+            packed = torch.cat([
+                sum_local, sqsum_local,
+                torch.full((1,), float(count_local), dtype=x.dtype,
+                           device=x.device)])
+            dist.all_reduce(packed, op=dist.ReduceOp.SUM,
+                            group=process_group)
+            c = x.shape[1]
+            sum_g, sqsum_g = packed[:c], packed[c:2 * c]
+            count_g = float(packed[2 * c])
+        else:
+            sum_g, sqsum_g, count_g = sum_local, sqsum_local, \
+                float(count_local)
+        mean = sum_g / count_g
+        var = sqsum_g / count_g - mean * mean
+        invstd = torch.rsqrt(var.clamp(min=0) + eps)
+        shape = [1, x.shape[1]] + [1] * (x.dim() - 2)
+        xhat = (x - mean.view(shape)) * invstd.view(shape)
+        out = xhat * weight.view(shape) + bias.view(shape)
+        ctx.save_for_backward(xhat, weight, invstd)
+        ctx.process_group = process_group
+        ctx.world_size = world_size
+        ctx.count_g = count_g
+        return out, mean.detach(), var.detach(), torch.tensor(count_g)
+
+    @staticmethod
+    def backward(ctx, grad_out, _gm, _gv, _gc):
+        xhat, weight, invstd = ctx.saved_tensors
+        reduce_dims = _flatten_to_nc(grad_out)
+        sum_dy = grad_out.sum(dim=reduce_dims)
+        sum_dy_xhat = (grad_out * xhat).sum(dim=reduce_dims)
+        # weight/bias grads: LOCAL sums (DDP averages them afterwards, the
+        # same contract as nn.SyncBatchNorm)
+        grad_weight = sum_dy_xhat.clone()
+        grad_bias = sum_dy.clone()
+        if ctx.world_size > 1:
+            c = sum_dy.shape[0]
+            packed = torch.cat([sum_dy, sum_dy_xhat])
+            dist.all_reduce(packed, op=dist.ReduceOp.SUM,
+                            group=ctx.process_group)
+            sum_dy, sum_dy_xhat = packed[:c], packed[c:]
+        count = ctx.count_g
+        shape = [1, grad_out.shape[1]] + [1] * (grad_out.dim() - 2)
+        grad_x = (weight * invstd).view(shape) * (
+            grad_out - (sum_dy / count).view(shape)
+            - xhat * (sum_dy_xhat / count).view(shape))
+        return grad_x, grad_weight, grad_bias, None, None, None
+
+
+class SyncBatchNorm(nn.Module):
+    """Drop-in for BatchNorm1d/2d with cross-replica statistics."""
+
+    def __init__(self, num_features: int, eps: float = 1e-5,
+                 momentum: float = 0.1, affine: bool = True,
+                 track_running_stats: bool = True,
+                 process_group=None):
+        super().__init__()
+        assert affine and track_running_stats, \
+            "BYOL uses affine+tracked BN only"
+        self.num_features = num_features
+        self.eps = eps
+        self.momentum = momentum
+        self.process_group = process_group
+        self.weight = nn.Parameter(torch.ones(num_features))
+        self.bias = nn.Parameter(torch.zeros(num_features))
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+        self.register_buffer("num_batches_tracked",
+                             torch.tensor(0, dtype=torch.long))
+
+    def extra_repr(self):
+        return f"{self.num_features}, eps={self.eps}, momentum={self.momentum}"
+
+    def _world(self) -> int:
+        if dist.is_available() and dist.is_initialized():
+            return dist.get_world_size(self.process_group)
+        return 1
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.dim() < 2:
+            raise ValueError("expected at least 2-D input")
+        if not self.training:
+            shape = [1, self.num_features] + [1] * (x.dim() - 2)
+            invstd = torch.rsqrt(self.running_var + self.eps)
+            return ((x - self.running_mean.view(shape))
+                    * (invstd * self.weight).view(shape)
+                    + self.bias.view(shape))
+        world = self._world()
+        out, mean, var, count = _SyncBNFunction.apply(
+            x, self.weight, self.bias, self.eps, self.process_group, world)
+        with torch.no_grad():
+            self.num_batches_tracked += 1
+            n = float(count)
+            unbiased = var * (n / max(n - 1.0, 1.0))
+            self.running_mean.mul_(1 - self.momentum).add_(
+                mean, alpha=self.momentum)
+            self.running_var.mul_(1 - self.momentum).add_(
+                unbiased, alpha=self.momentum)
+        return out
+
+
+def convert_sync_batchnorm(module: nn.Module,
+                           process_group=None) -> nn.Module:
+    """Replace every BatchNorm1d/2d/3d with SyncBatchNorm, copying state
+    (same contract as ``nn.SyncBatchNorm.convert_sync_batchnorm``)."""
+    if isinstance(module, nn.modules.batchnorm._BatchNorm):
+        sync = SyncBatchNorm(module.num_features, eps=module.eps,
+                             momentum=module.momentum,
+                             process_group=process_group)
+        with torch.no_grad():
+            sync.weight.copy_(module.weight)
+            sync.bias.copy_(module.bias)
+            sync.running_mean.copy_(module.running_mean)
+            sync.running_var.copy_(module.running_var)
+            sync.num_batches_tracked.copy_(module.num_batches_tracked)
+        return sync
+    for name, child in module.named_children():
+        module.add_module(name, convert_sync_batchnorm(child, process_group))
+    return module
