@@ -19,7 +19,14 @@ overlap, fused LARS+momentum step.  Nothing is skipped or cached.
 import argparse
 import json
 import os
+import sys
 import time
+
+# MIOpen immediate/fast find: the exhaustive find that
+# torch.backends.cudnn.benchmark=True requests costs many minutes per fresh
+# box (no tuning DB persists there).  FAST picks algos heuristically at
+# ~zero startup cost; override with MIOPEN_FIND_MODE/--miopen-benchmark.
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
 
 import torch
 import torch.distributed as dist
@@ -40,6 +47,8 @@ def parse():
     p.add_argument("--sync-bn", action="store_true", default=False)
     p.add_argument("--channels-last", action="store_true", default=False)
     p.add_argument("--bucket-cap-mb", type=float, default=32.0)
+    p.add_argument("--miopen-benchmark", action="store_true", default=False,
+                   help="exhaustive MIOpen find (slow first run)")
     return p.parse_args()
 
 
@@ -63,7 +72,7 @@ def main():
     from byol_amd.parallel.sync_bn import convert_sync_batchnorm
 
     torch.manual_seed(1234 + rank)
-    torch.backends.cudnn.benchmark = True
+    torch.backends.cudnn.benchmark = args.miopen_benchmark
 
     repr_size = {"resnet18": 512, "resnet34": 512}.get(args.arch, 2048)
     model = BYOL(arch=args.arch, base_network_output_size=repr_size,
@@ -126,6 +135,9 @@ def main():
 
     for i in range(args.warmup):
         step(i)
+        if rank == 0:
+            print(f"warmup {i + 1}/{args.warmup} done", file=sys.stderr,
+                  flush=True)
 
     if distributed:
         dist.barrier()

