@@ -22,11 +22,21 @@ import os
 import sys
 import time
 
-# MIOpen immediate/fast find: the exhaustive find that
-# torch.backends.cudnn.benchmark=True requests costs many minutes per fresh
-# box (no tuning DB persists there).  FAST picks algos heuristically at
-# ~zero startup cost; override with MIOPEN_FIND_MODE/--miopen-benchmark.
-os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+# MIOpen find setup: NORMAL find with the pre-tuned user DB shipped in the
+# repo (byol_amd/ops/miopen_udb, measured on MI355X) gives tuned solver
+# choices at near-zero startup cost on a fresh box.  The DB dir must be
+# writable (MIOpen appends new finds), so copy it aside.
+os.environ.setdefault("MIOPEN_FIND_MODE", "NORMAL")
+if "MIOPEN_USER_DB_PATH" not in os.environ:
+    import shutil
+    import tempfile
+    _src = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "byol_amd", "ops", "miopen_udb")
+    _dst = os.path.join(tempfile.gettempdir(),
+                        f"miopen_udb_{os.environ.get('LOCAL_RANK', 0)}")
+    if os.path.isdir(_src):
+        shutil.copytree(_src, _dst, dirs_exist_ok=True)
+        os.environ["MIOPEN_USER_DB_PATH"] = _dst
 
 import torch
 import torch.distributed as dist
@@ -45,7 +55,12 @@ def parse():
     p.add_argument("--dtype", type=str, default="fp32",
                    choices=["fp32", "bf16"])
     p.add_argument("--sync-bn", action="store_true", default=False)
-    p.add_argument("--channels-last", action="store_true", default=False)
+    p.add_argument("--channels-last", dest="channels_last",
+                   action="store_true", default=True,
+                   help="NHWC memory format (default on: the tuned MIOpen "
+                        "NHWC solvers are the fast fp32 path on gfx950)")
+    p.add_argument("--no-channels-last", dest="channels_last",
+                   action="store_false")
     p.add_argument("--bucket-cap-mb", type=float, default=32.0)
     p.add_argument("--miopen-benchmark", action="store_true", default=False,
                    help="exhaustive MIOpen find (slow first run)")
