@@ -161,11 +161,15 @@ class CosEMA(nn.Module):
 
 
 def _mlp_head(in_dim: int, latent_dim: int, out_dim: int) -> nn.Sequential:
-    # projector/predictor shape of the reference (/root/reference/main.py:194-205)
+    # projector/predictor shape of the reference
+    # (/root/reference/main.py:194-205).  The BN1d carries the ReLU fused
+    # into its HIP kernel; the Identity keeps the reference's Sequential
+    # indices (head.0 linear, head.1 bn, head.3 linear).
+    from ..ops.bn import FusedBatchNorm
     return nn.Sequential(
         nn.Linear(in_dim, latent_dim),
-        nn.BatchNorm1d(latent_dim),
-        nn.ReLU(),
+        FusedBatchNorm(latent_dim, relu=True),
+        nn.Identity(),
         nn.Linear(latent_dim, out_dim),
     )
 

@@ -2,22 +2,24 @@
 
 Replicates the architecture the reference reaches through torchvision
 (``/root/reference/main.py:190-193`` builds ``nn.Sequential(*children()[:-1])``
-of a torchvision ResNet, i.e. the full network minus the final fc but keeping
-the global average pool).  We build exactly that encoder directly — there is no
-torchvision in this environment, and owning the module tree lets us fuse
-conv+BN+ReLU blocks into HIP kernels later without fighting a vendored model.
+of a torchvision ResNet: the full network minus the final fc, keeping the
+global average pool).  Built directly — no torchvision here — and shaped for
+the MI355X hot path: every BatchNorm is a :class:`FusedBatchNorm` with the
+following ReLU (and, at block exits, the residual add) folded into the same
+HIP kernel, so a bottleneck block runs conv -> ONE fused bn+relu kernel pair
+instead of conv -> BN stats -> BN norm -> ReLU -> add (see
+``byol_amd/ops/bn.py``).  Numerics are identical to the torchvision module
+tree (same stride/width/downsample placement, kaiming init, BN eps/momentum).
 
-Supported archs: resnet18/34/50/101/152/200 (the reference exposes every
-torchvision arch; the ResNet family is the one its README and defaults use).
-
-Defaults (stride/width/downsample placement, kaiming init, BN eps/momentum)
-match torchvision's ResNet so numerics line up with the reference oracle.
+Supported archs: resnet18/34/50/101/152/200.
 """
 
 from typing import List, Optional, Type, Union
 
 import torch
 import torch.nn as nn
+
+from ..ops.bn import FusedBatchNorm
 
 __all__ = ["ResNetEncoder", "build_encoder", "ARCH_SPECS", "arch_names"]
 
@@ -32,6 +34,18 @@ def conv1x1(in_planes: int, out_planes: int, stride: int = 1) -> nn.Conv2d:
                      bias=False)
 
 
+class Downsample(nn.Module):
+    """1x1 strided conv + (non-relu) BN on the identity branch."""
+
+    def __init__(self, inplanes: int, outplanes: int, stride: int):
+        super().__init__()
+        self.conv = conv1x1(inplanes, outplanes, stride)
+        self.bn = FusedBatchNorm(outplanes)
+
+    def forward(self, x):
+        return self.bn(self.conv(x))
+
+
 class BasicBlock(nn.Module):
     expansion = 1
 
@@ -39,21 +53,16 @@ class BasicBlock(nn.Module):
                  downsample: Optional[nn.Module] = None):
         super().__init__()
         self.conv1 = conv3x3(inplanes, planes, stride)
-        self.bn1 = nn.BatchNorm2d(planes)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn1 = FusedBatchNorm(planes, relu=True)
         self.conv2 = conv3x3(planes, planes)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = FusedBatchNorm(planes, relu=True)  # fused add+relu
         self.downsample = downsample
         self.stride = stride
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
-        if self.downsample is not None:
-            identity = self.downsample(x)
-        out = out + identity
-        return self.relu(out)
+        identity = x if self.downsample is None else self.downsample(x)
+        out = self.bn1(self.conv1(x))
+        return self.bn2(self.conv2(out), residual=identity)
 
 
 class Bottleneck(nn.Module):
@@ -63,24 +72,20 @@ class Bottleneck(nn.Module):
                  downsample: Optional[nn.Module] = None):
         super().__init__()
         self.conv1 = conv1x1(inplanes, planes)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = FusedBatchNorm(planes, relu=True)
         self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = FusedBatchNorm(planes, relu=True)
         self.conv3 = conv1x1(planes, planes * self.expansion)
-        self.bn3 = nn.BatchNorm2d(planes * self.expansion)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn3 = FusedBatchNorm(planes * self.expansion,
+                                  relu=True)  # fused add+relu
         self.downsample = downsample
         self.stride = stride
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
-        if self.downsample is not None:
-            identity = self.downsample(x)
-        out = out + identity
-        return self.relu(out)
+        identity = x if self.downsample is None else self.downsample(x)
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        return self.bn3(self.conv3(out), residual=identity)
 
 
 ARCH_SPECS = {
@@ -100,9 +105,9 @@ def arch_names() -> List[str]:
 class ResNetEncoder(nn.Module):
     """ResNet minus the final fc: stem -> 4 stages -> global avgpool.
 
-    Output is ``(B, out_channels, 1, 1)``; callers ``.view(-1, C)`` it, the
-    same contract as the reference's ``children()[:-1]`` Sequential
-    (``/root/reference/main.py:190-193, 238``).
+    Output is ``(B, out_channels, 1, 1)``; callers ``.view(-1, C)`` it — the
+    reference's ``children()[:-1]`` contract
+    (``/root/reference/main.py:190-193,238``).
     """
 
     def __init__(self, block: Type[Union[BasicBlock, Bottleneck]],
@@ -111,8 +116,7 @@ class ResNetEncoder(nn.Module):
         self.inplanes = 64
         self.conv1 = nn.Conv2d(in_channels, 64, kernel_size=7, stride=2,
                                padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn1 = FusedBatchNorm(64, relu=True)
         self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
         self.layer1 = self._make_layer(block, 64, layers[0])
         self.layer2 = self._make_layer(block, 128, layers[1], stride=2)
@@ -125,7 +129,7 @@ class ResNetEncoder(nn.Module):
             if isinstance(m, nn.Conv2d):
                 nn.init.kaiming_normal_(m.weight, mode="fan_out",
                                         nonlinearity="relu")
-            elif isinstance(m, nn.BatchNorm2d):
+            elif isinstance(m, FusedBatchNorm):
                 nn.init.constant_(m.weight, 1)
                 nn.init.constant_(m.bias, 0)
 
@@ -133,17 +137,15 @@ class ResNetEncoder(nn.Module):
                     stride: int = 1) -> nn.Sequential:
         downsample = None
         if stride != 1 or self.inplanes != planes * block.expansion:
-            downsample = nn.Sequential(
-                conv1x1(self.inplanes, planes * block.expansion, stride),
-                nn.BatchNorm2d(planes * block.expansion),
-            )
+            downsample = Downsample(self.inplanes,
+                                    planes * block.expansion, stride)
         layers = [block(self.inplanes, planes, stride, downsample)]
         self.inplanes = planes * block.expansion
         layers += [block(self.inplanes, planes) for _ in range(1, blocks)]
         return nn.Sequential(*layers)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.relu(self.bn1(self.conv1(x)))
+        x = self.bn1(self.conv1(x))
         x = self.maxpool(x)
         x = self.layer1(x)
         x = self.layer2(x)

@@ -15,6 +15,24 @@ void launch_byol_loss_backward(const float* p1, const float* p2,
                                const float* stats, const float* grad_out,
                                float* g1, float* g2, int64_t n,
                                int64_t batch, hipStream_t stream);
+void launch_bn_stats(const float* x, float* acc, int64_t m, int c,
+                     hipStream_t stream);
+void launch_bn_finalize(const float* acc, float* mean, float* invstd,
+                        float* running_mean, float* running_var, float count,
+                        float eps, float momentum, int c, int update_running,
+                        hipStream_t stream);
+void launch_bn_apply(const float* x, const float* residual, const float* mean,
+                     const float* invstd, const float* weight,
+                     const float* bias, float* y, int64_t m, int c, int relu,
+                     hipStream_t stream);
+void launch_bn_bwd_reduce(const float* dy, const float* y, const float* x,
+                          const float* mean, const float* invstd, float* red,
+                          int64_t m, int c, int relu, hipStream_t stream);
+void launch_bn_bwd_apply(const float* dy, const float* y, const float* x,
+                         const float* mean, const float* invstd,
+                         const float* weight, const float* red, float* dx,
+                         float* dresidual, float inv_count, int64_t m, int c,
+                         int relu, hipStream_t stream);
 void launch_lars_momentum_step(float* p, const float* g, float* m,
                                float* norm_acc, float* alr,
                                const int64_t* seg_off,
@@ -105,9 +123,79 @@ void lars_momentum_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
       static_cast<int>(m_init), stream);
 }
 
+// --- fused BN(+residual)(+ReLU), NHWC/2-D rows-x-channels fp32 -------------
+// All tensors are raw [M, C]-layout fp32 views prepared by the Python
+// wrapper (byol_amd/ops/bn.py), which owns layout checks and SyncBN comm.
+
+void bn_stats(torch::Tensor x, torch::Tensor acc, int64_t m, int64_t c) {
+  CHECK_IN(x); CHECK_IN(acc);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_stats(x.data_ptr<float>(), acc.data_ptr<float>(), m, (int)c,
+                  stream);
+}
+
+void bn_finalize(torch::Tensor acc, torch::Tensor mean, torch::Tensor invstd,
+                 torch::Tensor running_mean, torch::Tensor running_var,
+                 double count, double eps, double momentum, int64_t c,
+                 int64_t update_running) {
+  CHECK_IN(acc); CHECK_IN(mean); CHECK_IN(invstd);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_finalize(acc.data_ptr<float>(), mean.data_ptr<float>(),
+                     invstd.data_ptr<float>(),
+                     update_running ? running_mean.data_ptr<float>() : nullptr,
+                     update_running ? running_var.data_ptr<float>() : nullptr,
+                     (float)count, (float)eps, (float)momentum, (int)c,
+                     (int)update_running, stream);
+}
+
+void bn_apply(torch::Tensor x, c10::optional<torch::Tensor> residual,
+              torch::Tensor mean, torch::Tensor invstd, torch::Tensor weight,
+              torch::Tensor bias, torch::Tensor y, int64_t m, int64_t c,
+              int64_t relu) {
+  CHECK_IN(x); CHECK_IN(mean); CHECK_IN(invstd); CHECK_IN(weight);
+  CHECK_IN(bias); CHECK_IN(y);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_apply(x.data_ptr<float>(),
+                  residual ? residual->data_ptr<float>() : nullptr,
+                  mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                  weight.data_ptr<float>(), bias.data_ptr<float>(),
+                  y.data_ptr<float>(), m, (int)c, (int)relu, stream);
+}
+
+void bn_bwd_reduce(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+                   torch::Tensor mean, torch::Tensor invstd,
+                   torch::Tensor red, int64_t m, int64_t c, int64_t relu) {
+  CHECK_IN(dy); CHECK_IN(y); CHECK_IN(x); CHECK_IN(red);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_bwd_reduce(dy.data_ptr<float>(), y.data_ptr<float>(),
+                       x.data_ptr<float>(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), red.data_ptr<float>(), m,
+                       (int)c, (int)relu, stream);
+}
+
+void bn_bwd_apply(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+                  torch::Tensor mean, torch::Tensor invstd,
+                  torch::Tensor weight, torch::Tensor red, torch::Tensor dx,
+                  c10::optional<torch::Tensor> dresidual, double inv_count,
+                  int64_t m, int64_t c, int64_t relu) {
+  CHECK_IN(dy); CHECK_IN(y); CHECK_IN(x); CHECK_IN(dx);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_bwd_apply(dy.data_ptr<float>(), y.data_ptr<float>(),
+                      x.data_ptr<float>(), mean.data_ptr<float>(),
+                      invstd.data_ptr<float>(), weight.data_ptr<float>(),
+                      red.data_ptr<float>(), dx.data_ptr<float>(),
+                      dresidual ? dresidual->data_ptr<float>() : nullptr,
+                      (float)inv_count, m, (int)c, (int)relu, stream);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("bn_stats", &bn_stats);
+  mod.def("bn_finalize", &bn_finalize);
+  mod.def("bn_apply", &bn_apply);
+  mod.def("bn_bwd_reduce", &bn_bwd_reduce);
+  mod.def("bn_bwd_apply", &bn_bwd_apply);
   mod.def("flat_ema_update", &flat_ema_update,
           "fused flat-parameter EMA update (gfx950)");
   mod.def("byol_loss_forward", &byol_loss_forward,

@@ -1,0 +1,366 @@
+#include "hip/hip_runtime.h"
+// Fused BatchNorm(+residual)(+ReLU) for NHWC fp32 on gfx950.
+//
+// Replaces the reference's per-BN kernel parade (MIOpen SpatialMeanVariance +
+// SpatialNorm + ATen ReLU + ATen residual add; backward dScaleDBias + dX +
+// ReLU-bwd + add-bwd — SURVEY.md K3/K6) with:
+//   fwd:  bn_stats (one pass, per-channel sum/sumsq, atomics)
+//         bn_finalize (tiny: mean/invstd + running-stat update)
+//         bn_apply (normalize + affine + residual-add + ReLU, one pass)
+//   bwd:  bn_bwd_reduce (dy_eff = relu-masked dy; per-channel sum_dy,
+//         sum_dy_xhat -> also dbias/dweight)
+//         bn_bwd_apply (dx and optional dresidual, one pass)
+//
+// Layout contract: x is [M rows][C channels] with C contiguous — i.e. torch
+// channels_last 4-D tensors (NHWC) or plain-contiguous 2-D [B, C] tensors
+// (the projector/predictor BN1d).  C must be a multiple of 4 (float4 path).
+//
+// SyncBN: the host wrapper all-reduces the packed stats buffer between
+// bn_stats and bn_finalize (and between bn_bwd_reduce and bn_bwd_apply),
+// giving cross-replica batch statistics with ONE small RCCL message per
+// layer.
+#include "common.h"
+
+// how many float4 channel-quads one thread owns at most: supports C<=8192
+#define MAX_Q 8
+
+struct Quad {
+  float4 s;
+  float4 ss;
+};
+
+// ---------------------------------------------------------------------------
+// stats: acc[0:C] = sum, acc[C:2C] = sum of squares
+// ---------------------------------------------------------------------------
+__global__ void bn_stats_kernel(const float* __restrict__ x,
+                                float* __restrict__ acc,
+                                int64_t m, int c) {
+  const int c4 = c >> 2;
+  const int nthread = blockDim.x;
+  // RPB row-lanes of CW channel-threads each (CW = min(c4, nthread))
+  const int cw = c4 < nthread ? c4 : nthread;
+  const int rpb = nthread / cw;           // rows processed in parallel
+  const int tc = threadIdx.x % cw;        // channel-thread index
+  const int tr = threadIdx.x / cw;        // row lane
+  const int kq = (c4 + cw - 1) / cw;      // quads per thread
+
+  float4 s[MAX_Q], ss[MAX_Q];
+  #pragma unroll
+  for (int k = 0; k < MAX_Q; ++k) {
+    s[k] = make_float4(0.f, 0.f, 0.f, 0.f);
+    ss[k] = make_float4(0.f, 0.f, 0.f, 0.f);
+  }
+
+  const float4* x4 = reinterpret_cast<const float4*>(x);
+  const int64_t row_stride = (int64_t)rpb * gridDim.x;
+  for (int64_t row = (int64_t)blockIdx.x * rpb + tr; row < m;
+       row += row_stride) {
+    if (tr < rpb) {
+      const int64_t base = row * c4;
+      for (int k = 0; k < kq; ++k) {
+        const int q = tc + k * cw;
+        if (q < c4) {
+          const float4 v = x4[base + q];
+          s[k].x += v.x; s[k].y += v.y; s[k].z += v.z; s[k].w += v.w;
+          ss[k] = make_float4(fmaf(v.x, v.x, ss[k].x),
+                              fmaf(v.y, v.y, ss[k].y),
+                              fmaf(v.z, v.z, ss[k].z),
+                              fmaf(v.w, v.w, ss[k].w));
+        }
+      }
+    }
+  }
+
+  // reduce across row lanes (same tc) through LDS, then atomics
+  __shared__ Quad scratch[256];
+  for (int k = 0; k < kq; ++k) {
+    scratch[threadIdx.x].s = s[k];
+    scratch[threadIdx.x].ss = ss[k];
+    __syncthreads();
+    if (tr == 0) {
+      float4 ts = s[k], tss = ss[k];
+      for (int r = 1; r < rpb; ++r) {
+        const Quad& o = scratch[tc + r * cw];
+        ts.x += o.s.x; ts.y += o.s.y; ts.z += o.s.z; ts.w += o.s.w;
+        tss.x += o.ss.x; tss.y += o.ss.y; tss.z += o.ss.z; tss.w += o.ss.w;
+      }
+      const int q = tc + k * cw;
+      if (q < c4) {
+        const int ch = q * 4;
+        atomicAdd(&acc[ch + 0], ts.x);
+        atomicAdd(&acc[ch + 1], ts.y);
+        atomicAdd(&acc[ch + 2], ts.z);
+        atomicAdd(&acc[ch + 3], ts.w);
+        atomicAdd(&acc[c + ch + 0], tss.x);
+        atomicAdd(&acc[c + ch + 1], tss.y);
+        atomicAdd(&acc[c + ch + 2], tss.z);
+        atomicAdd(&acc[c + ch + 3], tss.w);
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// mean/invstd from (possibly all-reduced) acc; update running stats
+__global__ void bn_finalize_kernel(const float* __restrict__ acc,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ invstd,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var,
+                                   float count, float eps, float momentum,
+                                   int c, int update_running) {
+  for (int ch = blockIdx.x * blockDim.x + threadIdx.x; ch < c;
+       ch += gridDim.x * blockDim.x) {
+    const float mu = acc[ch] / count;
+    float var = acc[c + ch] / count - mu * mu;
+    var = var < 0.f ? 0.f : var;
+    mean[ch] = mu;
+    invstd[ch] = rsqrtf(var + eps);
+    if (update_running) {
+      const float unbiased = var * (count / fmaxf(count - 1.f, 1.f));
+      running_mean[ch] = fmaf(momentum, mu - running_mean[ch],
+                              running_mean[ch]);
+      running_var[ch] = fmaf(momentum, unbiased - running_var[ch],
+                             running_var[ch]);
+    }
+  }
+}
+
+// y = relu?(w*(x-mean)*invstd + b [+ residual])
+__global__ void bn_apply_kernel(const float* __restrict__ x,
+                                const float* __restrict__ residual,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ invstd,
+                                const float* __restrict__ weight,
+                                const float* __restrict__ bias,
+                                float* __restrict__ y,
+                                int64_t m, int c, int relu) {
+  const int c4 = c >> 2;
+  const int64_t n4 = m * c4;
+  const float4* x4 = reinterpret_cast<const float4*>(x);
+  const float4* r4 = reinterpret_cast<const float4*>(residual);
+  const float4* mean4 = reinterpret_cast<const float4*>(mean);
+  const float4* inv4 = reinterpret_cast<const float4*>(invstd);
+  const float4* w4 = reinterpret_cast<const float4*>(weight);
+  const float4* b4 = reinterpret_cast<const float4*>(bias);
+  float4* y4 = reinterpret_cast<float4*>(y);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    const int q = (int)(i % c4);
+    const float4 v = x4[i];
+    const float4 mu = mean4[q], is = inv4[q], w = w4[q], b = b4[q];
+    float4 o;
+    o.x = fmaf((v.x - mu.x) * is.x, w.x, b.x);
+    o.y = fmaf((v.y - mu.y) * is.y, w.y, b.y);
+    o.z = fmaf((v.z - mu.z) * is.z, w.z, b.z);
+    o.w = fmaf((v.w - mu.w) * is.w, w.w, b.w);
+    if (residual != nullptr) {
+      const float4 r = r4[i];
+      o.x += r.x; o.y += r.y; o.z += r.z; o.w += r.w;
+    }
+    if (relu) {
+      o.x = fmaxf(o.x, 0.f); o.y = fmaxf(o.y, 0.f);
+      o.z = fmaxf(o.z, 0.f); o.w = fmaxf(o.w, 0.f);
+    }
+    y4[i] = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward
+// red[0:C] = sum(dy_eff), red[C:2C] = sum(dy_eff * xhat)
+// dy_eff = relu? (y > 0 ? dy : 0) : dy       (xhat from x, mean, invstd)
+// ---------------------------------------------------------------------------
+__global__ void bn_bwd_reduce_kernel(const float* __restrict__ dy,
+                                     const float* __restrict__ y,
+                                     const float* __restrict__ x,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     float* __restrict__ red,
+                                     int64_t m, int c, int relu) {
+  const int c4 = c >> 2;
+  const int nthread = blockDim.x;
+  const int cw = c4 < nthread ? c4 : nthread;
+  const int rpb = nthread / cw;
+  const int tc = threadIdx.x % cw;
+  const int tr = threadIdx.x / cw;
+  const int kq = (c4 + cw - 1) / cw;
+
+  float4 s1[MAX_Q], s2[MAX_Q];
+  #pragma unroll
+  for (int k = 0; k < MAX_Q; ++k) {
+    s1[k] = make_float4(0.f, 0.f, 0.f, 0.f);
+    s2[k] = make_float4(0.f, 0.f, 0.f, 0.f);
+  }
+  const float4* dy4 = reinterpret_cast<const float4*>(dy);
+  const float4* y4 = reinterpret_cast<const float4*>(y);
+  const float4* x4 = reinterpret_cast<const float4*>(x);
+  const float4* mean4 = reinterpret_cast<const float4*>(mean);
+  const float4* inv4 = reinterpret_cast<const float4*>(invstd);
+  const int64_t row_stride = (int64_t)rpb * gridDim.x;
+  for (int64_t row = (int64_t)blockIdx.x * rpb + tr; row < m;
+       row += row_stride) {
+    if (tr < rpb) {
+      const int64_t base = row * c4;
+      for (int k = 0; k < kq; ++k) {
+        const int q = tc + k * cw;
+        if (q < c4) {
+          float4 g = dy4[base + q];
+          if (relu) {
+            const float4 yy = y4[base + q];
+            g.x = yy.x > 0.f ? g.x : 0.f;
+            g.y = yy.y > 0.f ? g.y : 0.f;
+            g.z = yy.z > 0.f ? g.z : 0.f;
+            g.w = yy.w > 0.f ? g.w : 0.f;
+          }
+          const float4 v = x4[base + q];
+          const float4 mu = mean4[q], is = inv4[q];
+          s1[k].x += g.x; s1[k].y += g.y; s1[k].z += g.z; s1[k].w += g.w;
+          s2[k].x = fmaf(g.x, (v.x - mu.x) * is.x, s2[k].x);
+          s2[k].y = fmaf(g.y, (v.y - mu.y) * is.y, s2[k].y);
+          s2[k].z = fmaf(g.z, (v.z - mu.z) * is.z, s2[k].z);
+          s2[k].w = fmaf(g.w, (v.w - mu.w) * is.w, s2[k].w);
+        }
+      }
+    }
+  }
+
+  __shared__ Quad scratch[256];
+  for (int k = 0; k < kq; ++k) {
+    scratch[threadIdx.x].s = s1[k];
+    scratch[threadIdx.x].ss = s2[k];
+    __syncthreads();
+    if (tr == 0) {
+      float4 t1 = s1[k], t2 = s2[k];
+      for (int r = 1; r < rpb; ++r) {
+        const Quad& o = scratch[tc + r * cw];
+        t1.x += o.s.x; t1.y += o.s.y; t1.z += o.s.z; t1.w += o.s.w;
+        t2.x += o.ss.x; t2.y += o.ss.y; t2.z += o.ss.z; t2.w += o.ss.w;
+      }
+      const int q = tc + k * cw;
+      if (q < c4) {
+        const int ch = q * 4;
+        atomicAdd(&red[ch + 0], t1.x);
+        atomicAdd(&red[ch + 1], t1.y);
+        atomicAdd(&red[ch + 2], t1.z);
+        atomicAdd(&red[ch + 3], t1.w);
+        atomicAdd(&red[c + ch + 0], t2.x);
+        atomicAdd(&red[c + ch + 1], t2.y);
+        atomicAdd(&red[c + ch + 2], t2.z);
+        atomicAdd(&red[c + ch + 3], t2.w);
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// dx = (w*invstd) * (dy_eff - red1/M - xhat * red2/M); dresidual = dy_eff
+__global__ void bn_bwd_apply_kernel(const float* __restrict__ dy,
+                                    const float* __restrict__ y,
+                                    const float* __restrict__ x,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    const float* __restrict__ weight,
+                                    const float* __restrict__ red,
+                                    float* __restrict__ dx,
+                                    float* __restrict__ dresidual,
+                                    float inv_count, int64_t m, int c,
+                                    int relu) {
+  const int c4 = c >> 2;
+  const int64_t n4 = m * c4;
+  const float4* dy4 = reinterpret_cast<const float4*>(dy);
+  const float4* y4 = reinterpret_cast<const float4*>(y);
+  const float4* x4 = reinterpret_cast<const float4*>(x);
+  const float4* mean4 = reinterpret_cast<const float4*>(mean);
+  const float4* inv4 = reinterpret_cast<const float4*>(invstd);
+  const float4* w4 = reinterpret_cast<const float4*>(weight);
+  const float4* r1 = reinterpret_cast<const float4*>(red);
+  const float4* r2 = reinterpret_cast<const float4*>(red + c);
+  float4* dx4 = reinterpret_cast<float4*>(dx);
+  float4* dr4 = reinterpret_cast<float4*>(dresidual);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    const int q = (int)(i % c4);
+    float4 g = dy4[i];
+    if (relu) {
+      const float4 yy = y4[i];
+      g.x = yy.x > 0.f ? g.x : 0.f;
+      g.y = yy.y > 0.f ? g.y : 0.f;
+      g.z = yy.z > 0.f ? g.z : 0.f;
+      g.w = yy.w > 0.f ? g.w : 0.f;
+    }
+    if (dresidual != nullptr) dr4[i] = g;
+    const float4 v = x4[i];
+    const float4 mu = mean4[q], is = inv4[q], w = w4[q];
+    const float4 m1 = r1[q], m2 = r2[q];
+    float4 o;
+    o.x = w.x * is.x * (g.x - m1.x * inv_count
+                        - (v.x - mu.x) * is.x * m2.x * inv_count);
+    o.y = w.y * is.y * (g.y - m1.y * inv_count
+                        - (v.y - mu.y) * is.y * m2.y * inv_count);
+    o.z = w.z * is.z * (g.z - m1.z * inv_count
+                        - (v.z - mu.z) * is.z * m2.z * inv_count);
+    o.w = w.w * is.w * (g.w - m1.w * inv_count
+                        - (v.w - mu.w) * is.w * m2.w * inv_count);
+    dx4[i] = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+static int stats_grid(int64_t m, int c) {
+  // enough blocks to fill 256 CUs x 8 XCDs comfortably, capped
+  const int c4 = c >> 2;
+  const int cw = c4 < 256 ? c4 : 256;
+  const int rpb = 256 / cw;
+  int64_t g = (m + rpb - 1) / rpb;
+  if (g > 2048) g = 2048;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+void launch_bn_stats(const float* x, float* acc, int64_t m, int c,
+                     hipStream_t stream) {
+  hipLaunchKernelGGL(bn_stats_kernel, dim3(stats_grid(m, c)), dim3(256), 0,
+                     stream, x, acc, m, c);
+}
+
+void launch_bn_finalize(const float* acc, float* mean, float* invstd,
+                        float* running_mean, float* running_var, float count,
+                        float eps, float momentum, int c, int update_running,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((c + 255) / 256), dim3(256), 0,
+                     stream, acc, mean, invstd, running_mean, running_var,
+                     count, eps, momentum, c, update_running);
+}
+
+void launch_bn_apply(const float* x, const float* residual, const float* mean,
+                     const float* invstd, const float* weight,
+                     const float* bias, float* y, int64_t m, int c, int relu,
+                     hipStream_t stream) {
+  const int64_t n4 = m * (c >> 2);
+  hipLaunchKernelGGL(bn_apply_kernel, dim3(grid_1d(n4, 256)), dim3(256), 0,
+                     stream, x, residual, mean, invstd, weight, bias, y, m, c,
+                     relu);
+}
+
+void launch_bn_bwd_reduce(const float* dy, const float* y, const float* x,
+                          const float* mean, const float* invstd, float* red,
+                          int64_t m, int c, int relu, hipStream_t stream) {
+  hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(stats_grid(m, c)), dim3(256),
+                     0, stream, dy, y, x, mean, invstd, red, m, c, relu);
+}
+
+void launch_bn_bwd_apply(const float* dy, const float* y, const float* x,
+                         const float* mean, const float* invstd,
+                         const float* weight, const float* red, float* dx,
+                         float* dresidual, float inv_count, int64_t m, int c,
+                         int relu, hipStream_t stream) {
+  const int64_t n4 = m * (c >> 2);
+  hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3(grid_1d(n4, 256)), dim3(256),
+                     0, stream, dy, y, x, mean, invstd, weight, red, dx,
+                     dresidual, inv_count, m, c, relu);
+}

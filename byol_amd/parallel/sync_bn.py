@@ -137,8 +137,15 @@ class SyncBatchNorm(nn.Module):
 
 def convert_sync_batchnorm(module: nn.Module,
                            process_group=None) -> nn.Module:
-    """Replace every BatchNorm1d/2d/3d with SyncBatchNorm, copying state
-    (same contract as ``nn.SyncBatchNorm.convert_sync_batchnorm``)."""
+    """Enable cross-replica statistics on every norm layer (the contract of
+    ``nn.SyncBatchNorm.convert_sync_batchnorm``).  FusedBatchNorm modules
+    flip their ``sync`` flag in place (the fused HIP kernels already split
+    at the stats all-reduce point); plain torch BatchNorms are replaced."""
+    from ..ops.bn import FusedBatchNorm
+    if isinstance(module, FusedBatchNorm):
+        module.sync = True
+        module.process_group = process_group
+        return module
     if isinstance(module, nn.modules.batchnorm._BatchNorm):
         sync = SyncBatchNorm(module.num_features, eps=module.eps,
                              momentum=module.momentum,

@@ -19,6 +19,7 @@ sources = [
     os.path.join(CSRC, "ema.hip"),
     os.path.join(CSRC, "byol_loss.hip"),
     os.path.join(CSRC, "lars.hip"),
+    os.path.join(CSRC, "bn_fused.hip"),
 ]
 
 setup(

@@ -165,6 +165,86 @@ def test_byol_fused_loss_used_in_model_path(ext):
     assert loss.grad_fn.__class__.__name__.startswith("_FusedBYOLLoss")
 
 
+@pytest.mark.parametrize("shape,relu,residual", [
+    ((16, 64, 14, 14), False, False),
+    ((16, 64, 14, 14), True, False),
+    ((8, 256, 7, 7), True, True),
+    ((32, 2048, 7, 7), True, True),
+])
+def test_fused_bn2d_hip_vs_oracle(ext, shape, relu, residual):
+    """HIP NHWC fused BN vs torch BatchNorm2d oracle (fwd+bwd+running)."""
+    import torch.nn as nn
+    from byol_amd.ops.bn import FusedBatchNorm
+    torch.manual_seed(5)
+    c = shape[1]
+    fused = FusedBatchNorm(c, relu=relu).cuda()
+    ref = nn.BatchNorm2d(c).cuda()
+    x1 = torch.randn(shape, device="cuda").to(
+        memory_format=torch.channels_last).requires_grad_(True)
+    x2 = x1.detach().clone().requires_grad_(True)
+    r1 = r2 = None
+    if residual:
+        r1 = torch.randn(shape, device="cuda").to(
+            memory_format=torch.channels_last).requires_grad_(True)
+        r2 = r1.detach().clone().requires_grad_(True)
+    y1 = fused(x1, residual=r1)
+    y2 = ref(x2)
+    if residual:
+        y2 = y2 + r2
+    if relu:
+        y2 = torch.relu(y2)
+    assert torch.allclose(y1, y2, rtol=1e-4, atol=1e-5), \
+        (y1 - y2).abs().max().item()
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g)
+    assert torch.allclose(x1.grad, x2.grad, rtol=1e-4, atol=1e-4), \
+        (x1.grad - x2.grad).abs().max().item()
+    if residual:
+        assert torch.allclose(r1.grad, r2.grad, rtol=1e-4, atol=1e-5)
+    assert torch.allclose(fused.weight.grad, ref.weight.grad, rtol=1e-3,
+                          atol=1e-3)
+    assert torch.allclose(fused.bias.grad, ref.bias.grad, rtol=1e-3,
+                          atol=1e-3)
+    assert torch.allclose(fused.running_mean, ref.running_mean, atol=1e-4)
+    assert torch.allclose(fused.running_var, ref.running_var, rtol=1e-3,
+                          atol=1e-4)
+
+
+def test_fused_bn1d_hip_vs_oracle(ext):
+    import torch.nn as nn
+    from byol_amd.ops.bn import FusedBatchNorm
+    torch.manual_seed(6)
+    fused = FusedBatchNorm(4096, relu=True).cuda()
+    ref = nn.BatchNorm1d(4096).cuda()
+    x1 = torch.randn(512, 4096, device="cuda", requires_grad=True)
+    x2 = x1.detach().clone().requires_grad_(True)
+    y1 = fused(x1)
+    y2 = torch.relu(ref(x2))
+    assert torch.allclose(y1, y2, rtol=1e-4, atol=1e-5)
+    y1.sum().backward()
+    y2.sum().backward()
+    assert torch.allclose(x1.grad, x2.grad, rtol=1e-4, atol=1e-4)
+
+
+def test_fused_bn_eval_mode_hip(ext):
+    import torch.nn as nn
+    from byol_amd.ops.bn import FusedBatchNorm
+    torch.manual_seed(7)
+    fused = FusedBatchNorm(64).cuda()
+    ref = nn.BatchNorm2d(64).cuda()
+    x = torch.randn(8, 64, 7, 7, device="cuda").to(
+        memory_format=torch.channels_last)
+    fused(x)
+    ref(x)
+    fused.eval()
+    ref.eval()
+    with torch.no_grad():
+        x2 = torch.randn(4, 64, 7, 7, device="cuda").to(
+            memory_format=torch.channels_last)
+        assert torch.allclose(fused(x2), ref(x2), rtol=1e-4, atol=1e-5)
+
+
 def test_smoke_entrypoint():
     import __graft_entry__
     __graft_entry__.smoke()
