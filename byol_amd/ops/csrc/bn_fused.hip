@@ -31,6 +31,7 @@ struct Quad {
 // ---------------------------------------------------------------------------
 // stats: acc[0:C] = sum, acc[C:2C] = sum of squares
 // ---------------------------------------------------------------------------
+template <int KQ>
 __global__ void bn_stats_kernel(const float* __restrict__ x,
                                 float* __restrict__ acc,
                                 int64_t m, int c) {
@@ -41,22 +42,24 @@ __global__ void bn_stats_kernel(const float* __restrict__ x,
   const int rpb = nthread / cw;           // rows processed in parallel
   const int tc = threadIdx.x % cw;        // channel-thread index
   const int tr = threadIdx.x / cw;        // row lane
-  const int kq = (c4 + cw - 1) / cw;      // quads per thread
+  const bool active = tr < rpb;
 
-  float4 s[MAX_Q], ss[MAX_Q];
+  // KQ is a compile-time bound so the accumulators stay in VGPRs
+  float4 s[KQ], ss[KQ];
   #pragma unroll
-  for (int k = 0; k < MAX_Q; ++k) {
+  for (int k = 0; k < KQ; ++k) {
     s[k] = make_float4(0.f, 0.f, 0.f, 0.f);
     ss[k] = make_float4(0.f, 0.f, 0.f, 0.f);
   }
 
   const float4* x4 = reinterpret_cast<const float4*>(x);
   const int64_t row_stride = (int64_t)rpb * gridDim.x;
-  for (int64_t row = (int64_t)blockIdx.x * rpb + tr; row < m;
-       row += row_stride) {
-    if (tr < rpb) {
+  if (active) {
+    for (int64_t row = (int64_t)blockIdx.x * rpb + tr; row < m;
+         row += row_stride) {
       const int64_t base = row * c4;
-      for (int k = 0; k < kq; ++k) {
+      #pragma unroll
+      for (int k = 0; k < KQ; ++k) {
         const int q = tc + k * cw;
         if (q < c4) {
           const float4 v = x4[base + q];
@@ -72,7 +75,8 @@ __global__ void bn_stats_kernel(const float* __restrict__ x,
 
   // reduce across row lanes (same tc) through LDS, then atomics
   __shared__ Quad scratch[256];
-  for (int k = 0; k < kq; ++k) {
+  #pragma unroll
+  for (int k = 0; k < KQ; ++k) {
     scratch[threadIdx.x].s = s[k];
     scratch[threadIdx.x].ss = ss[k];
     __syncthreads();
@@ -135,6 +139,7 @@ __global__ void bn_apply_kernel(const float* __restrict__ x,
                                 float* __restrict__ y,
                                 int64_t m, int c, int relu) {
   const int c4 = c >> 2;
+  const int q_mask = ((c4 & (c4 - 1)) == 0) ? (c4 - 1) : -1;
   const int64_t n4 = m * c4;
   const float4* x4 = reinterpret_cast<const float4*>(x);
   const float4* r4 = reinterpret_cast<const float4*>(residual);
@@ -146,7 +151,7 @@ __global__ void bn_apply_kernel(const float* __restrict__ x,
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
        i += stride) {
-    const int q = (int)(i % c4);
+    const int q = q_mask >= 0 ? (int)(i & q_mask) : (int)(i % c4);
     const float4 v = x4[i];
     const float4 mu = mean4[q], is = inv4[q], w = w4[q], b = b4[q];
     float4 o;
@@ -171,6 +176,7 @@ __global__ void bn_apply_kernel(const float* __restrict__ x,
 // red[0:C] = sum(dy_eff), red[C:2C] = sum(dy_eff * xhat)
 // dy_eff = relu? (y > 0 ? dy : 0) : dy       (xhat from x, mean, invstd)
 // ---------------------------------------------------------------------------
+template <int KQ>
 __global__ void bn_bwd_reduce_kernel(const float* __restrict__ dy,
                                      const float* __restrict__ y,
                                      const float* __restrict__ x,
@@ -184,11 +190,11 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ dy,
   const int rpb = nthread / cw;
   const int tc = threadIdx.x % cw;
   const int tr = threadIdx.x / cw;
-  const int kq = (c4 + cw - 1) / cw;
+  const bool active = tr < rpb;
 
-  float4 s1[MAX_Q], s2[MAX_Q];
+  float4 s1[KQ], s2[KQ];
   #pragma unroll
-  for (int k = 0; k < MAX_Q; ++k) {
+  for (int k = 0; k < KQ; ++k) {
     s1[k] = make_float4(0.f, 0.f, 0.f, 0.f);
     s2[k] = make_float4(0.f, 0.f, 0.f, 0.f);
   }
@@ -198,11 +204,12 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ dy,
   const float4* mean4 = reinterpret_cast<const float4*>(mean);
   const float4* inv4 = reinterpret_cast<const float4*>(invstd);
   const int64_t row_stride = (int64_t)rpb * gridDim.x;
-  for (int64_t row = (int64_t)blockIdx.x * rpb + tr; row < m;
-       row += row_stride) {
-    if (tr < rpb) {
+  if (active) {
+    for (int64_t row = (int64_t)blockIdx.x * rpb + tr; row < m;
+         row += row_stride) {
       const int64_t base = row * c4;
-      for (int k = 0; k < kq; ++k) {
+      #pragma unroll
+      for (int k = 0; k < KQ; ++k) {
         const int q = tc + k * cw;
         if (q < c4) {
           float4 g = dy4[base + q];
@@ -226,7 +233,8 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ dy,
   }
 
   __shared__ Quad scratch[256];
-  for (int k = 0; k < kq; ++k) {
+  #pragma unroll
+  for (int k = 0; k < KQ; ++k) {
     scratch[threadIdx.x].s = s1[k];
     scratch[threadIdx.x].ss = s2[k];
     __syncthreads();
@@ -267,6 +275,7 @@ __global__ void bn_bwd_apply_kernel(const float* __restrict__ dy,
                                     float inv_count, int64_t m, int c,
                                     int relu) {
   const int c4 = c >> 2;
+  const int q_mask = ((c4 & (c4 - 1)) == 0) ? (c4 - 1) : -1;
   const int64_t n4 = m * c4;
   const float4* dy4 = reinterpret_cast<const float4*>(dy);
   const float4* y4 = reinterpret_cast<const float4*>(y);
@@ -281,7 +290,7 @@ __global__ void bn_bwd_apply_kernel(const float* __restrict__ dy,
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
        i += stride) {
-    const int q = (int)(i % c4);
+    const int q = q_mask >= 0 ? (int)(i & q_mask) : (int)(i % c4);
     float4 g = dy4[i];
     if (relu) {
       const float4 yy = y4[i];
@@ -321,10 +330,29 @@ static int stats_grid(int64_t m, int c) {
   return (int)g;
 }
 
+static int kq_for(int c) {
+  const int c4 = c >> 2;
+  const int cw = c4 < 256 ? c4 : 256;
+  const int kq = (c4 + cw - 1) / cw;
+  if (kq <= 1) return 1;
+  if (kq <= 2) return 2;
+  if (kq <= 4) return 4;
+  return 8;
+}
+
 void launch_bn_stats(const float* x, float* acc, int64_t m, int c,
                      hipStream_t stream) {
-  hipLaunchKernelGGL(bn_stats_kernel, dim3(stats_grid(m, c)), dim3(256), 0,
-                     stream, x, acc, m, c);
+  const dim3 g(stats_grid(m, c)), b(256);
+  switch (kq_for(c)) {
+    case 1: hipLaunchKernelGGL(bn_stats_kernel<1>, g, b, 0, stream, x, acc,
+                               m, c); break;
+    case 2: hipLaunchKernelGGL(bn_stats_kernel<2>, g, b, 0, stream, x, acc,
+                               m, c); break;
+    case 4: hipLaunchKernelGGL(bn_stats_kernel<4>, g, b, 0, stream, x, acc,
+                               m, c); break;
+    default: hipLaunchKernelGGL(bn_stats_kernel<8>, g, b, 0, stream, x, acc,
+                                m, c); break;
+  }
 }
 
 void launch_bn_finalize(const float* acc, float* mean, float* invstd,
@@ -349,8 +377,17 @@ void launch_bn_apply(const float* x, const float* residual, const float* mean,
 void launch_bn_bwd_reduce(const float* dy, const float* y, const float* x,
                           const float* mean, const float* invstd, float* red,
                           int64_t m, int c, int relu, hipStream_t stream) {
-  hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(stats_grid(m, c)), dim3(256),
-                     0, stream, dy, y, x, mean, invstd, red, m, c, relu);
+  const dim3 g(stats_grid(m, c)), b(256);
+  switch (kq_for(c)) {
+    case 1: hipLaunchKernelGGL(bn_bwd_reduce_kernel<1>, g, b, 0, stream, dy,
+                               y, x, mean, invstd, red, m, c, relu); break;
+    case 2: hipLaunchKernelGGL(bn_bwd_reduce_kernel<2>, g, b, 0, stream, dy,
+                               y, x, mean, invstd, red, m, c, relu); break;
+    case 4: hipLaunchKernelGGL(bn_bwd_reduce_kernel<4>, g, b, 0, stream, dy,
+                               y, x, mean, invstd, red, m, c, relu); break;
+    default: hipLaunchKernelGGL(bn_bwd_reduce_kernel<8>, g, b, 0, stream, dy,
+                                y, x, mean, invstd, red, m, c, relu); break;
+  }
 }
 
 void launch_bn_bwd_apply(const float* dy, const float* y, const float* x,
