@@ -53,21 +53,29 @@ __global__ void bn_stats_kernel(const float* __restrict__ x,
   }
 
   const float4* x4 = reinterpret_cast<const float4*>(x);
-  const int64_t row_stride = (int64_t)rpb * gridDim.x;
+  // 4-row unroll: 4 independent 16-B loads in flight per thread (a single
+  // outstanding load leaves the kernel latency-bound at ~1.3 TB/s)
+  const int64_t row_stride = (int64_t)rpb * 4 * gridDim.x;
   if (active) {
-    for (int64_t row = (int64_t)blockIdx.x * rpb + tr; row < m;
+    for (int64_t row = (int64_t)blockIdx.x * rpb * 4 + tr; row < m;
          row += row_stride) {
-      const int64_t base = row * c4;
       #pragma unroll
-      for (int k = 0; k < KQ; ++k) {
-        const int q = tc + k * cw;
-        if (q < c4) {
-          const float4 v = x4[base + q];
-          s[k].x += v.x; s[k].y += v.y; s[k].z += v.z; s[k].w += v.w;
-          ss[k] = make_float4(fmaf(v.x, v.x, ss[k].x),
-                              fmaf(v.y, v.y, ss[k].y),
-                              fmaf(v.z, v.z, ss[k].z),
-                              fmaf(v.w, v.w, ss[k].w));
+      for (int j = 0; j < 4; ++j) {
+        const int64_t r = row + (int64_t)j * rpb;
+        if (r < m) {
+          const int64_t base = r * c4;
+          #pragma unroll
+          for (int k = 0; k < KQ; ++k) {
+            const int q = tc + k * cw;
+            if (q < c4) {
+              const float4 v = x4[base + q];
+              s[k].x += v.x; s[k].y += v.y; s[k].z += v.z; s[k].w += v.w;
+              ss[k] = make_float4(fmaf(v.x, v.x, ss[k].x),
+                                  fmaf(v.y, v.y, ss[k].y),
+                                  fmaf(v.z, v.z, ss[k].z),
+                                  fmaf(v.w, v.w, ss[k].w));
+            }
+          }
         }
       }
     }
@@ -203,11 +211,16 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ dy,
   const float4* x4 = reinterpret_cast<const float4*>(x);
   const float4* mean4 = reinterpret_cast<const float4*>(mean);
   const float4* inv4 = reinterpret_cast<const float4*>(invstd);
-  const int64_t row_stride = (int64_t)rpb * gridDim.x;
+  // 2-row unroll: 6 independent loads in flight (3 streams x 2 rows)
+  const int64_t row_stride = (int64_t)rpb * 2 * gridDim.x;
   if (active) {
-    for (int64_t row = (int64_t)blockIdx.x * rpb + tr; row < m;
+    for (int64_t row = (int64_t)blockIdx.x * rpb * 2 + tr; row < m;
          row += row_stride) {
-      const int64_t base = row * c4;
+      #pragma unroll
+      for (int j = 0; j < 2; ++j) {
+      const int64_t r = row + (int64_t)j * rpb;
+      if (r >= m) continue;
+      const int64_t base = r * c4;
       #pragma unroll
       for (int k = 0; k < KQ; ++k) {
         const int q = tc + k * cw;
@@ -228,6 +241,7 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ dy,
           s2[k].z = fmaf(g.z, (v.z - mu.z) * is.z, s2[k].z);
           s2[k].w = fmaf(g.w, (v.w - mu.w) * is.w, s2[k].w);
         }
+      }
       }
     }
   }
@@ -325,7 +339,7 @@ static int stats_grid(int64_t m, int c) {
   const int cw = c4 < 256 ? c4 : 256;
   const int rpb = 256 / cw;
   int64_t g = (m + rpb - 1) / rpb;
-  if (g > 2048) g = 2048;
+  if (g > 4096) g = 4096;
   if (g < 1) g = 1;
   return (int)g;
 }
