@@ -66,15 +66,19 @@ class _FusedBNFunction(torch.autograd.Function):
         world = _world(process_group) if sync else 1
 
         if training:
-            acc = torch.zeros(2 * c, device=x.device, dtype=torch.float32)
-            C.bn_stats(xv, acc, m, c)
+            nslots = 64  # atomic-contention fanout (see csrc/bn_fused.hip)
+            acc = torch.zeros(nslots * 2 * c, device=x.device,
+                              dtype=torch.float32)
+            C.bn_stats(xv, acc, m, c, nslots)
+            acc2c = torch.empty(2 * c, device=x.device, dtype=torch.float32)
+            C.bn_reduce_slots(acc, acc2c, nslots)
             count = float(m)
             if world > 1:
-                dist.all_reduce(acc, group=process_group)
+                dist.all_reduce(acc2c, group=process_group)
                 count = float(m * world)
             mean = torch.empty(c, device=x.device, dtype=torch.float32)
             invstd = torch.empty_like(mean)
-            C.bn_finalize(acc, mean, invstd, running_mean, running_var,
+            C.bn_finalize(acc2c, mean, invstd, running_mean, running_var,
                           count, eps, momentum, c,
                           1 if running_mean is not None else 0)
         else:
@@ -107,9 +111,13 @@ class _FusedBNFunction(torch.autograd.Function):
         xv = _rows_view(x, c)
         yv = _rows_view(y, c)
         m = xv.shape[0]
-        red = torch.zeros(2 * c, device=x.device, dtype=torch.float32)
-        C.bn_bwd_reduce(dyv, yv, xv, mean, invstd, red, m, c,
-                        1 if ctx.relu else 0)
+        nslots = 64
+        red_s = torch.zeros(nslots * 2 * c, device=x.device,
+                            dtype=torch.float32)
+        C.bn_bwd_reduce(dyv, yv, xv, mean, invstd, red_s, m, c,
+                        1 if ctx.relu else 0, nslots)
+        red = torch.empty(2 * c, device=x.device, dtype=torch.float32)
+        C.bn_reduce_slots(red_s, red, nslots)
         # weight/bias grads are LOCAL sums (DDP averages them afterwards,
         # same contract as nn.SyncBatchNorm)
         db = red[:c].clone()

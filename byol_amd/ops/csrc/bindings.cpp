@@ -16,7 +16,9 @@ void launch_byol_loss_backward(const float* p1, const float* p2,
                                float* g1, float* g2, int64_t n,
                                int64_t batch, hipStream_t stream);
 void launch_bn_stats(const float* x, float* acc, int64_t m, int c,
-                     hipStream_t stream);
+                     int slot_mask, hipStream_t stream);
+void launch_bn_reduce_slots(const float* in, float* out, int n2c, int nslots,
+                            hipStream_t stream);
 void launch_bn_finalize(const float* acc, float* mean, float* invstd,
                         float* running_mean, float* running_var, float count,
                         float eps, float momentum, int c, int update_running,
@@ -27,7 +29,8 @@ void launch_bn_apply(const float* x, const float* residual, const float* mean,
                      hipStream_t stream);
 void launch_bn_bwd_reduce(const float* dy, const float* y, const float* x,
                           const float* mean, const float* invstd, float* red,
-                          int64_t m, int c, int relu, hipStream_t stream);
+                          int64_t m, int c, int relu, int slot_mask,
+                          hipStream_t stream);
 void launch_bn_bwd_apply(const float* dy, const float* y, const float* x,
                          const float* mean, const float* invstd,
                          const float* weight, const float* red, float* dx,
@@ -127,11 +130,20 @@ void lars_momentum_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 // All tensors are raw [M, C]-layout fp32 views prepared by the Python
 // wrapper (byol_amd/ops/bn.py), which owns layout checks and SyncBN comm.
 
-void bn_stats(torch::Tensor x, torch::Tensor acc, int64_t m, int64_t c) {
+void bn_stats(torch::Tensor x, torch::Tensor acc, int64_t m, int64_t c,
+              int64_t nslots) {
   CHECK_IN(x); CHECK_IN(acc);
+  TORCH_CHECK(acc.numel() == nslots * 2 * c, "acc must be [nslots][2C]");
   auto stream = at::hip::getCurrentHIPStream();
   launch_bn_stats(x.data_ptr<float>(), acc.data_ptr<float>(), m, (int)c,
-                  stream);
+                  (int)(nslots - 1), stream);
+}
+
+void bn_reduce_slots(torch::Tensor in, torch::Tensor out, int64_t nslots) {
+  CHECK_IN(in); CHECK_IN(out);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_reduce_slots(in.data_ptr<float>(), out.data_ptr<float>(),
+                         (int)out.numel(), (int)nslots, stream);
 }
 
 void bn_finalize(torch::Tensor acc, torch::Tensor mean, torch::Tensor invstd,
@@ -164,13 +176,15 @@ void bn_apply(torch::Tensor x, c10::optional<torch::Tensor> residual,
 
 void bn_bwd_reduce(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
                    torch::Tensor mean, torch::Tensor invstd,
-                   torch::Tensor red, int64_t m, int64_t c, int64_t relu) {
+                   torch::Tensor red, int64_t m, int64_t c, int64_t relu,
+                   int64_t nslots) {
   CHECK_IN(dy); CHECK_IN(y); CHECK_IN(x); CHECK_IN(red);
+  TORCH_CHECK(red.numel() == nslots * 2 * c, "red must be [nslots][2C]");
   auto stream = at::hip::getCurrentHIPStream();
   launch_bn_bwd_reduce(dy.data_ptr<float>(), y.data_ptr<float>(),
                        x.data_ptr<float>(), mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), red.data_ptr<float>(), m,
-                       (int)c, (int)relu, stream);
+                       (int)c, (int)relu, (int)(nslots - 1), stream);
 }
 
 void bn_bwd_apply(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
@@ -192,6 +206,7 @@ void bn_bwd_apply(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_stats", &bn_stats);
+  mod.def("bn_reduce_slots", &bn_reduce_slots);
   mod.def("bn_finalize", &bn_finalize);
   mod.def("bn_apply", &bn_apply);
   mod.def("bn_bwd_reduce", &bn_bwd_reduce);

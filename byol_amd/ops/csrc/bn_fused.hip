@@ -34,7 +34,10 @@ struct Quad {
 template <int KQ>
 __global__ void bn_stats_kernel(const float* __restrict__ x,
                                 float* __restrict__ acc,
-                                int64_t m, int c) {
+                                int64_t m, int c, int slot_mask) {
+  // acc is [nslots][2C]; blocks hash into slots so per-address atomic
+  // contention is num_blocks/nslots instead of num_blocks
+  float* const acc_slot = acc + (int64_t)(blockIdx.x & slot_mask) * 2 * c;
   const int c4 = c >> 2;
   const int nthread = blockDim.x;
   // RPB row-lanes of CW channel-threads each (CW = min(c4, nthread))
@@ -98,17 +101,29 @@ __global__ void bn_stats_kernel(const float* __restrict__ x,
       const int q = tc + k * cw;
       if (q < c4) {
         const int ch = q * 4;
-        atomicAdd(&acc[ch + 0], ts.x);
-        atomicAdd(&acc[ch + 1], ts.y);
-        atomicAdd(&acc[ch + 2], ts.z);
-        atomicAdd(&acc[ch + 3], ts.w);
-        atomicAdd(&acc[c + ch + 0], tss.x);
-        atomicAdd(&acc[c + ch + 1], tss.y);
-        atomicAdd(&acc[c + ch + 2], tss.z);
-        atomicAdd(&acc[c + ch + 3], tss.w);
+        atomicAdd(&acc_slot[ch + 0], ts.x);
+        atomicAdd(&acc_slot[ch + 1], ts.y);
+        atomicAdd(&acc_slot[ch + 2], ts.z);
+        atomicAdd(&acc_slot[ch + 3], ts.w);
+        atomicAdd(&acc_slot[c + ch + 0], tss.x);
+        atomicAdd(&acc_slot[c + ch + 1], tss.y);
+        atomicAdd(&acc_slot[c + ch + 2], tss.z);
+        atomicAdd(&acc_slot[c + ch + 3], tss.w);
       }
     }
     __syncthreads();
+  }
+}
+
+// out[0:2C] = sum over slots of in[slot][0:2C]
+__global__ void bn_reduce_slots_kernel(const float* __restrict__ in,
+                                       float* __restrict__ out,
+                                       int n2c, int nslots) {
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < n2c;
+       i += gridDim.x * blockDim.x) {
+    float v = 0.f;
+    for (int sl = 0; sl < nslots; ++sl) v += in[(int64_t)sl * n2c + i];
+    out[i] = v;
   }
 }
 
@@ -191,7 +206,9 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ dy,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
                                      float* __restrict__ red,
-                                     int64_t m, int c, int relu) {
+                                     int64_t m, int c, int relu,
+                                     int slot_mask) {
+  float* const red_slot = red + (int64_t)(blockIdx.x & slot_mask) * 2 * c;
   const int c4 = c >> 2;
   const int nthread = blockDim.x;
   const int cw = c4 < nthread ? c4 : nthread;
@@ -262,14 +279,14 @@ __global__ void bn_bwd_reduce_kernel(const float* __restrict__ dy,
       const int q = tc + k * cw;
       if (q < c4) {
         const int ch = q * 4;
-        atomicAdd(&red[ch + 0], t1.x);
-        atomicAdd(&red[ch + 1], t1.y);
-        atomicAdd(&red[ch + 2], t1.z);
-        atomicAdd(&red[ch + 3], t1.w);
-        atomicAdd(&red[c + ch + 0], t2.x);
-        atomicAdd(&red[c + ch + 1], t2.y);
-        atomicAdd(&red[c + ch + 2], t2.z);
-        atomicAdd(&red[c + ch + 3], t2.w);
+        atomicAdd(&red_slot[ch + 0], t1.x);
+        atomicAdd(&red_slot[ch + 1], t1.y);
+        atomicAdd(&red_slot[ch + 2], t1.z);
+        atomicAdd(&red_slot[ch + 3], t1.w);
+        atomicAdd(&red_slot[c + ch + 0], t2.x);
+        atomicAdd(&red_slot[c + ch + 1], t2.y);
+        atomicAdd(&red_slot[c + ch + 2], t2.z);
+        atomicAdd(&red_slot[c + ch + 3], t2.w);
       }
     }
     __syncthreads();
@@ -355,18 +372,24 @@ static int kq_for(int c) {
 }
 
 void launch_bn_stats(const float* x, float* acc, int64_t m, int c,
-                     hipStream_t stream) {
+                     int slot_mask, hipStream_t stream) {
   const dim3 g(stats_grid(m, c)), b(256);
   switch (kq_for(c)) {
     case 1: hipLaunchKernelGGL(bn_stats_kernel<1>, g, b, 0, stream, x, acc,
-                               m, c); break;
+                               m, c, slot_mask); break;
     case 2: hipLaunchKernelGGL(bn_stats_kernel<2>, g, b, 0, stream, x, acc,
-                               m, c); break;
+                               m, c, slot_mask); break;
     case 4: hipLaunchKernelGGL(bn_stats_kernel<4>, g, b, 0, stream, x, acc,
-                               m, c); break;
+                               m, c, slot_mask); break;
     default: hipLaunchKernelGGL(bn_stats_kernel<8>, g, b, 0, stream, x, acc,
-                                m, c); break;
+                                m, c, slot_mask); break;
   }
+}
+
+void launch_bn_reduce_slots(const float* in, float* out, int n2c, int nslots,
+                            hipStream_t stream) {
+  hipLaunchKernelGGL(bn_reduce_slots_kernel, dim3((n2c + 255) / 256),
+                     dim3(256), 0, stream, in, out, n2c, nslots);
 }
 
 void launch_bn_finalize(const float* acc, float* mean, float* invstd,
@@ -390,17 +413,22 @@ void launch_bn_apply(const float* x, const float* residual, const float* mean,
 
 void launch_bn_bwd_reduce(const float* dy, const float* y, const float* x,
                           const float* mean, const float* invstd, float* red,
-                          int64_t m, int c, int relu, hipStream_t stream) {
+                          int64_t m, int c, int relu, int slot_mask,
+                          hipStream_t stream) {
   const dim3 g(stats_grid(m, c)), b(256);
   switch (kq_for(c)) {
     case 1: hipLaunchKernelGGL(bn_bwd_reduce_kernel<1>, g, b, 0, stream, dy,
-                               y, x, mean, invstd, red, m, c, relu); break;
+                               y, x, mean, invstd, red, m, c, relu,
+                               slot_mask); break;
     case 2: hipLaunchKernelGGL(bn_bwd_reduce_kernel<2>, g, b, 0, stream, dy,
-                               y, x, mean, invstd, red, m, c, relu); break;
+                               y, x, mean, invstd, red, m, c, relu,
+                               slot_mask); break;
     case 4: hipLaunchKernelGGL(bn_bwd_reduce_kernel<4>, g, b, 0, stream, dy,
-                               y, x, mean, invstd, red, m, c, relu); break;
+                               y, x, mean, invstd, red, m, c, relu,
+                               slot_mask); break;
     default: hipLaunchKernelGGL(bn_bwd_reduce_kernel<8>, g, b, 0, stream, dy,
-                                y, x, mean, invstd, red, m, c, relu); break;
+                                y, x, mean, invstd, red, m, c, relu,
+                                slot_mask); break;
   }
 }
 
