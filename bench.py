@@ -139,8 +139,11 @@ def main():
                 out["online_prediction2"].float(),
                 out["target_projection1"].float(),
                 out["target_projection2"].float())
-            loss = loss + F.cross_entropy(out["linear_preds"].float(),
-                                          torch.cat([lab, lab]))
+            from byol_amd.ops.classifier import cross_entropy_topk
+            ce, _a1, _a5 = cross_entropy_topk(
+                out["linear_preds"].float().contiguous(),
+                torch.cat([lab, lab]))
+            loss = loss + ce
         opt.zero_grad()
         loss.backward()
         if distributed:

@@ -130,6 +130,8 @@ def build_parser() -> argparse.ArgumentParser:
                         help="synthetic task: test set size")
     parser.add_argument("--channels-last", action="store_true", default=False,
                         help="NHWC memory format for conv throughput")
+    parser.add_argument("--perf-stats", action="store_true", default=False,
+                        help="per-phase HIP-event timing summary per epoch")
     return parser
 
 

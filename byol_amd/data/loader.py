@@ -118,6 +118,46 @@ class FolderImages(Dataset):
         return img, label
 
 
+class _SourceDataset(Dataset):
+    """Raw (image, label) pairs resized to a fixed source size — what the
+    GPU augmentation pipeline consumes (decode/resize on CPU workers, all
+    augmentation on device, like the reference's DALI path)."""
+
+    def __init__(self, base: Dataset, src_size: int):
+        from .transforms import Resize
+        self.base = base
+        self.resize = Resize((src_size, src_size))
+
+    def __len__(self):
+        return len(self.base)
+
+    def __getitem__(self, idx):
+        img, label = self.base[idx]
+        return self.resize(img), label
+
+
+class _GPUAugmentIterable:
+    """Wraps a raw-batch loader; yields (aug1, aug2, labels) on device."""
+
+    def __init__(self, raw_loader, out_size: int, jitter_strength: float,
+                 dali_mode: bool, seed: int):
+        from .gpu_augment import GPUTwoViewAugment
+        self.raw_loader = raw_loader
+        self.pipe = GPUTwoViewAugment(out_size,
+                                      jitter_strength=jitter_strength,
+                                      dali_mode=dali_mode, seed=seed)
+
+    def __len__(self):
+        return len(self.raw_loader)
+
+    def __iter__(self):
+        for imgs, labels in self.raw_loader:
+            imgs = imgs.cuda(non_blocking=True)
+            labels = labels.cuda(non_blocking=True)
+            aug1, aug2 = self.pipe(imgs)
+            yield aug1, aug2, labels
+
+
 class Loader:
     def __init__(self, train_dataset: Dataset, test_dataset: Dataset,
                  input_shape, output_size: int, batch_size: int,
@@ -178,12 +218,26 @@ def get_loader(train_transform=None, test_transform=None, **kwargs):
     else:
         raise ValueError(f"unknown task {task!r}")
 
-    train_ds = _TwoViewDataset(train_base, train_transform, test_transform,
-                               train=True)
+    gpu_augment = "dali" in task and cuda
+    if gpu_augment:
+        # DALI-parity path: CPU workers only decode+resize; crop/flip/
+        # jitter/gray run as HIP kernels on device (byol_amd/ops/csrc/
+        # augment.hip); blur on the torchvision-recipe tasks only.
+        src_size = max(image_size + image_size // 8, image_size + 8)
+        train_ds = _SourceDataset(train_base, src_size)
+    else:
+        train_ds = _TwoViewDataset(train_base, train_transform,
+                                   test_transform, train=True)
     test_ds = _TwoViewDataset(test_base, train_transform, test_transform,
                               train=False)
-    return Loader(train_ds, test_ds,
-                  input_shape=(3, image_size, image_size),
-                  output_size=output_size, batch_size=batch_size,
-                  num_replicas=num_replicas, rank=rank, workers=workers,
-                  seed=seed, pin_memory=cuda)
+    loader = Loader(train_ds, test_ds,
+                    input_shape=(3, image_size, image_size),
+                    output_size=output_size, batch_size=batch_size,
+                    num_replicas=num_replicas, rank=rank, workers=workers,
+                    seed=seed, pin_memory=cuda)
+    if gpu_augment:
+        loader.train_loader = _GPUAugmentIterable(
+            loader.train_loader, image_size,
+            jitter_strength=kwargs.get("color_jitter_strength", 1.0),
+            dali_mode=True, seed=seed)
+    return loader

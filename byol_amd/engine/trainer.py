@@ -64,6 +64,9 @@ def build_loader_model_grapher(args):
         network = convert_sync_batchnorm(network)
     if args.cuda:
         network = network.cuda()
+        if getattr(args, "channels_last", False):
+            # NHWC: the tuned MIOpen solvers + our fused BN kernels' layout
+            network.base_network.to(memory_format=torch.channels_last)
     network = layers.init_weights(network, init=args.weight_initialization)
     network.finalize()  # flat param space + EMA prime (terminal)
     lazy_generate_modules(network, loader.train_loader, args)
@@ -193,14 +196,20 @@ def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
     is_ddp = isinstance(model, FlatDDP)
     autocast_ctx = (torch.autocast("cuda", dtype=torch.bfloat16)
                     if args.half and args.cuda else _dummy_context())
+    from ..profiling import PhaseTimer
+    timer = PhaseTimer(enabled=getattr(args, "perf_stats", False),
+                       use_cuda=args.cuda)
 
+    timer.start("data")
     for num_minibatches, (augmentation1, augmentation2, labels) in \
             enumerate(loader):
         if args.cuda:
             augmentation1 = augmentation1.cuda(non_blocking=True)
             augmentation2 = augmentation2.cuda(non_blocking=True)
             labels = labels.cuda(non_blocking=True)
+        timer.stop()
 
+        timer.start("forward")
         with torch.no_grad() if is_eval else _dummy_context():
             with autocast_ctx:
                 if is_eval and args.polyak_ema > 0:
@@ -210,6 +219,8 @@ def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
                 else:
                     output_dict = model(augmentation1, augmentation2)
 
+                timer.stop()
+                timer.start("loss")
                 byol_loss = loss_function(
                     online_prediction1=output_dict["online_prediction1"].float(),
                     online_prediction2=output_dict["online_prediction2"].float(),
@@ -217,11 +228,12 @@ def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
                     target_projection2=output_dict["target_projection2"].float())
                 classifier_labels = labels if is_eval else \
                     torch.cat([labels, labels], 0)
-                classifier_loss = F.cross_entropy(
-                    input=output_dict["linear_preds"].float(),
-                    target=classifier_labels)
-            acc1, acc5 = metrics.topk(output=output_dict["linear_preds"],
-                                      target=classifier_labels, topk=(1, 5))
+                # fused CE + top-1/top-5 in one HIP kernel on GPU
+                # (byol_amd/ops/classifier.py); composed oracle elsewhere
+                from ..ops.classifier import cross_entropy_topk
+                classifier_loss, acc1, acc5 = cross_entropy_topk(
+                    output_dict["linear_preds"].float().contiguous(),
+                    classifier_labels)
             loss_t = {
                 "loss_mean": byol_loss + classifier_loss,
                 "byol_loss_mean": byol_loss,
@@ -234,21 +246,29 @@ def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
             loss_map = detached_t if not loss_map else \
                 _sum_scalars(loss_map, detached_t)
             num_samples += augmentation1.size(0)
+            timer.add_samples(augmentation1.size(0))
+        timer.stop()
 
         if not is_eval:
+            timer.start("backward")
             optimizer.zero_grad()
             loss_t["loss_mean"].backward()
             if is_ddp:
                 model.finish_grad_sync()
+            timer.stop()
+            timer.start("optimizer")
             if args.clip > 0:
                 nn.utils.clip_grad_value_(model.parameters(), args.clip)
             optimizer.step()
             if args.polyak_ema > 0:
                 layers.polyak_ema_parameters(model, args.polyak_ema)
+            timer.stop()
             del loss_t
 
         if args.debug_step:
             break
+        timer.start("data")
+    timer.stop()
 
     loss_map = {k: v / (num_minibatches + 1) for k, v in loss_map.items()}
 
@@ -260,6 +280,9 @@ def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
         float(loss_map["loss_mean"]),
         float(loss_map["top1_mean"]),
         float(loss_map["top5_mean"])))
+
+    if getattr(args, "perf_stats", False):
+        print(timer.epoch_summary())
 
     register_plots(dict(loss_map), grapher, epoch=epoch, args=args,
                    prefix=prefix)

@@ -36,6 +36,17 @@ void launch_bn_bwd_apply(const float* dy, const float* y, const float* x,
                          const float* weight, const float* red, float* dx,
                          float* dresidual, float inv_count, int64_t m, int c,
                          int relu, hipStream_t stream);
+void launch_aug_sample(const float* src, float* dst, float* gray_sum,
+                       const float* crop, int b, int hs, int ws, int s,
+                       hipStream_t stream);
+void launch_aug_color(float* img, const float* gray_sum, const float* cparam,
+                      int b, int s, hipStream_t stream);
+void launch_ce_topk_fwd(const float* logits, const int64_t* labels,
+                        float* out, float* row_stats, int m, int n,
+                        hipStream_t stream);
+void launch_ce_bwd(const float* logits, const int64_t* labels,
+                   const float* row_stats, const float* grad_out,
+                   float* dlogits, int m, int n, hipStream_t stream);
 void launch_lars_momentum_step(float* p, const float* g, float* m,
                                float* norm_acc, float* alr,
                                const int64_t* seg_off,
@@ -202,9 +213,53 @@ void bn_bwd_apply(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
                       (float)inv_count, m, (int)c, (int)relu, stream);
 }
 
+void aug_sample(torch::Tensor src, torch::Tensor dst, torch::Tensor gray_sum,
+                torch::Tensor crop, int64_t hs, int64_t ws, int64_t s) {
+  CHECK_IN(src); CHECK_IN(dst); CHECK_IN(gray_sum); CHECK_IN(crop);
+  const int b = gray_sum.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_aug_sample(src.data_ptr<float>(), dst.data_ptr<float>(),
+                    gray_sum.data_ptr<float>(), crop.data_ptr<float>(), b,
+                    (int)hs, (int)ws, (int)s, stream);
+}
+
+void aug_color(torch::Tensor img, torch::Tensor gray_sum,
+               torch::Tensor cparam, int64_t s) {
+  CHECK_IN(img); CHECK_IN(gray_sum); CHECK_IN(cparam);
+  const int b = gray_sum.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_aug_color(img.data_ptr<float>(), gray_sum.data_ptr<float>(),
+                   cparam.data_ptr<float>(), b, (int)s, stream);
+}
+
+void ce_topk_fwd(torch::Tensor logits, torch::Tensor labels,
+                 torch::Tensor out, torch::Tensor row_stats) {
+  CHECK_IN(logits); CHECK_IN(out); CHECK_IN(row_stats);
+  TORCH_CHECK(labels.scalar_type() == at::kLong, "labels must be int64");
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_ce_topk_fwd(logits.data_ptr<float>(), labels.data_ptr<int64_t>(),
+                     out.data_ptr<float>(), row_stats.data_ptr<float>(),
+                     (int)logits.size(0), (int)logits.size(1), stream);
+}
+
+void ce_bwd(torch::Tensor logits, torch::Tensor labels,
+            torch::Tensor row_stats, torch::Tensor grad_out,
+            torch::Tensor dlogits) {
+  CHECK_IN(logits); CHECK_IN(row_stats); CHECK_IN(dlogits);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_ce_bwd(logits.data_ptr<float>(), labels.data_ptr<int64_t>(),
+                row_stats.data_ptr<float>(), grad_out.data_ptr<float>(),
+                dlogits.data_ptr<float>(), (int)logits.size(0),
+                (int)logits.size(1), stream);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("ce_topk_fwd", &ce_topk_fwd);
+  mod.def("ce_bwd", &ce_bwd);
+  mod.def("aug_sample", &aug_sample);
+  mod.def("aug_color", &aug_color);
   mod.def("bn_stats", &bn_stats);
   mod.def("bn_reduce_slots", &bn_reduce_slots);
   mod.def("bn_finalize", &bn_finalize);

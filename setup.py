@@ -20,6 +20,8 @@ sources = [
     os.path.join(CSRC, "byol_loss.hip"),
     os.path.join(CSRC, "lars.hip"),
     os.path.join(CSRC, "bn_fused.hip"),
+    os.path.join(CSRC, "augment.hip"),
+    os.path.join(CSRC, "cetopk.hip"),
 ]
 
 setup(

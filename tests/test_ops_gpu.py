@@ -245,6 +245,90 @@ def test_fused_bn_eval_mode_hip(ext):
         assert torch.allclose(fused(x2), ref(x2), rtol=1e-4, atol=1e-5)
 
 
+def test_fused_ce_topk_vs_oracle(ext):
+    import torch.nn.functional as F
+    from byol_amd.engine import metrics
+    from byol_amd.ops.classifier import _FusedCE
+    torch.manual_seed(8)
+    m, n = 1024, 1000
+    logits = torch.randn(m, n, device="cuda", requires_grad=True)
+    labels = torch.randint(n, (m,), device="cuda")
+    loss, acc1, acc5 = _FusedCE.apply(logits, labels)
+    ref_loss = F.cross_entropy(logits.detach(), labels)
+    r1, r5 = metrics.topk(logits.detach(), labels, topk=(1, 5))
+    assert torch.allclose(loss, ref_loss, rtol=1e-5, atol=1e-6)
+    assert torch.allclose(acc1, r1, atol=1e-3), (float(acc1), float(r1))
+    assert torch.allclose(acc5, r5, atol=1e-3), (float(acc5), float(r5))
+    loss.backward()
+    g_fused = logits.grad.clone()
+    logits2 = logits.detach().clone().requires_grad_(True)
+    F.cross_entropy(logits2, labels).backward()
+    assert torch.allclose(g_fused, logits2.grad, rtol=1e-4, atol=1e-7), \
+        (g_fused - logits2.grad).abs().max().item()
+
+
+def test_aug_sample_matches_interpolate(ext):
+    """Identity crop + no flip must equal F.interpolate bilinear resize."""
+    import torch.nn.functional as F
+    torch.manual_seed(9)
+    b, hs, ws, s = 4, 64, 48, 32
+    src = torch.rand(b, 3, hs, ws, device="cuda")
+    src_nhwc = src.permute(0, 2, 3, 1).contiguous()
+    crop = torch.tensor([[0.0, 0.0, hs, ws, 0.0]] * b,
+                        device="cuda").reshape(-1)
+    dst = torch.empty(b, s, s, 3, device="cuda")
+    gray = torch.zeros(b, device="cuda")
+    ext.aug_sample(src_nhwc.reshape(-1), dst.reshape(-1), gray, crop,
+                   hs, ws, s)
+    want = F.interpolate(src, size=(s, s), mode="bilinear",
+                         align_corners=False)
+    got = dst.permute(0, 3, 1, 2)
+    assert torch.allclose(got, want, atol=1e-5), \
+        (got - want).abs().max().item()
+    # gray accumulator = sum of gray over output pixels
+    gwant = (0.299 * want[:, 0] + 0.587 * want[:, 1]
+             + 0.114 * want[:, 2]).sum(dim=(1, 2))
+    assert torch.allclose(gray, gwant, rtol=1e-3, atol=1e-2)
+
+
+def test_aug_color_matches_reference(ext):
+    from byol_amd.data.gpu_augment import apply_color_reference
+    torch.manual_seed(10)
+    b, s = 6, 16
+    img = torch.rand(b, s, s, 3, device="cuda")
+    cparam = torch.zeros(b, 10, device="cuda")
+    for i in range(b):
+        cparam[i] = torch.tensor(
+            [1.0, 0.7 + 0.1 * i, 1.3 - 0.05 * i, 0.9, 0.05 * (i - 3),
+             1.0 if i == 5 else 0.0,
+             *torch.randperm(4, generator=torch.Generator().manual_seed(i)
+                             ).tolist()])
+    gray = (0.299 * img[..., 0] + 0.587 * img[..., 1]
+            + 0.114 * img[..., 2]).sum(dim=(1, 2))
+    gray_mean = gray / (s * s)
+    work = img.clone()
+    ext.aug_color(work.reshape(-1), gray, cparam.reshape(-1).contiguous(), s)
+    want = apply_color_reference(img.permute(0, 3, 1, 2), cparam.cpu(),
+                                 gray_mean)
+    got = work.permute(0, 3, 1, 2)
+    assert torch.allclose(got, want, atol=1e-4), \
+        (got - want).abs().max().item()
+
+
+def test_gpu_two_view_pipeline_end_to_end(ext):
+    from byol_amd.data.gpu_augment import GPUTwoViewAugment
+    torch.manual_seed(11)
+    pipe = GPUTwoViewAugment(out_size=32, jitter_strength=1.0,
+                             dali_mode=False, seed=3)
+    batch = torch.rand(8, 3, 40, 40, device="cuda")
+    a1, a2 = pipe(batch)
+    for a in (a1, a2):
+        assert a.shape == (8, 3, 32, 32)
+        assert a.is_contiguous(memory_format=torch.channels_last)
+        assert float(a.min()) >= 0.0 and float(a.max()) <= 1.0
+    assert not torch.equal(a1, a2)
+
+
 def test_smoke_entrypoint():
     import __graft_entry__
     __graft_entry__.smoke()
