@@ -30,6 +30,10 @@ def conv3x3(in_planes: int, out_planes: int, stride: int = 1) -> nn.Conv2d:
 
 
 def conv1x1(in_planes: int, out_planes: int, stride: int = 1) -> nn.Conv2d:
+    if stride == 1 and in_planes % 32 == 0:
+        # hand-written f32 MFMA GEMM path with MIOpen fallback dispatch
+        from ..ops.conv import MFMAConv1x1
+        return MFMAConv1x1(in_planes, out_planes, kernel_size=1, bias=False)
     return nn.Conv2d(in_planes, out_planes, kernel_size=1, stride=stride,
                      bias=False)
 

@@ -1,0 +1,92 @@
+"""MFMA 1x1 convolution (f32, NHWC) — hand-written gfx950 kernels with
+per-shape dispatch against MIOpen.
+
+A stride-1 1x1 conv in NHWC is exactly a GEMM over M = N*H*W rows;
+``csrc/conv1x1.hip`` implements fwd/dgrad/wgrad on the f32-input MFMA
+(``v_mfma_f32_32x32x2_f32``, the exact-f32 matrix path — 155 TF measured on
+this chip).  ``tools/conv_microbench.py`` measures these against MIOpen's
+igemm solvers shape-by-shape; dispatch is controlled by:
+
+* env ``BYOL_MFMA_CONV1X1`` = "0" (off), "1" (always), "auto" (default:
+  on for the shapes the microbench showed at parity-or-better).
+"""
+
+import os
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from . import extension, has_extension
+
+__all__ = ["MFMAConv1x1", "mfma_conv_mode"]
+
+
+def mfma_conv_mode() -> str:
+    return os.environ.get("BYOL_MFMA_CONV1X1", "auto")
+
+
+# Shapes (K=Cin, N=Cout) where the MFMA kernel measured >= MIOpen on
+# MI355X at bs-512-class M (see profiles/ microbench results); "auto" mode
+# enables exactly these.  Updated from tools/conv_microbench.py runs.
+_AUTO_SHAPES = set()
+
+
+def _eligible(x: torch.Tensor, weight: torch.Tensor) -> bool:
+    if not (x.is_cuda and x.dtype == torch.float32 and has_extension()):
+        return False
+    if not x.is_contiguous(memory_format=torch.channels_last):
+        return False
+    k = weight.shape[1]
+    if k % 32 != 0:
+        return False
+    mode = mfma_conv_mode()
+    if mode == "0":
+        return False
+    if mode == "1":
+        return True
+    return (k, weight.shape[0]) in _AUTO_SHAPES
+
+
+def _rows(t: torch.Tensor, c: int) -> torch.Tensor:
+    return t.permute(0, 2, 3, 1).reshape(-1, c)
+
+
+class _Conv1x1Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight):
+        C = extension()
+        b, k, h, w = x.shape
+        n = weight.shape[0]
+        m = b * h * w
+        wv = weight.reshape(n, k).contiguous()
+        y = torch.empty(b, n, h, w, device=x.device, dtype=x.dtype) \
+            .to(memory_format=torch.channels_last)
+        C.conv1x1_fwd(_rows(x, k), wv, _rows(y, n), m, k, n)
+        ctx.save_for_backward(x, wv)
+        ctx.dims = (b, k, h, w, n, m)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = extension()
+        x, wv = ctx.saved_tensors
+        b, k, h, w, n, m = ctx.dims
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx = torch.empty_like(x)
+        C.conv1x1_dgrad(_rows(dy, n), wv, _rows(dx, k), m, n, k)
+        dw = torch.zeros(n, k, device=x.device, dtype=x.dtype)
+        C.conv1x1_wgrad(_rows(dy, n), _rows(x, k), dw, m, n, k)
+        return dx, dw.view(n, k, 1, 1)
+
+
+class MFMAConv1x1(nn.Conv2d):
+    """Drop-in for stride-1 bias-free 1x1 nn.Conv2d; routes to the MFMA
+    kernels when eligible, MIOpen otherwise."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if _eligible(x, self.weight):
+            return _Conv1x1Fn.apply(x, self.weight)
+        return F.conv2d(x, self.weight, self.bias, self.stride,
+                        self.padding, self.dilation, self.groups)

@@ -47,6 +47,12 @@ void launch_ce_topk_fwd(const float* logits, const int64_t* labels,
 void launch_ce_bwd(const float* logits, const int64_t* labels,
                    const float* row_stats, const float* grad_out,
                    float* dlogits, int m, int n, hipStream_t stream);
+void launch_conv1x1_fwd(const float* x, const float* w, float* y, int64_t m,
+                        int k, int n, hipStream_t stream);
+void launch_conv1x1_dgrad(const float* dy, const float* w, float* dx,
+                          int64_t m, int n, int k, hipStream_t stream);
+void launch_conv1x1_wgrad(const float* dy, const float* x, float* dw,
+                          int64_t m, int n, int k, hipStream_t stream);
 void launch_lars_momentum_step(float* p, const float* g, float* m,
                                float* norm_acc, float* alr,
                                const int64_t* seg_off,
@@ -253,9 +259,36 @@ void ce_bwd(torch::Tensor logits, torch::Tensor labels,
                 (int)logits.size(1), stream);
 }
 
+void conv1x1_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor y,
+                 int64_t m, int64_t k, int64_t n) {
+  CHECK_IN(x); CHECK_IN(w); CHECK_IN(y);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_conv1x1_fwd(x.data_ptr<float>(), w.data_ptr<float>(),
+                     y.data_ptr<float>(), m, (int)k, (int)n, stream);
+}
+
+void conv1x1_dgrad(torch::Tensor dy, torch::Tensor w, torch::Tensor dx,
+                   int64_t m, int64_t n, int64_t k) {
+  CHECK_IN(dy); CHECK_IN(w); CHECK_IN(dx);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_conv1x1_dgrad(dy.data_ptr<float>(), w.data_ptr<float>(),
+                       dx.data_ptr<float>(), m, (int)n, (int)k, stream);
+}
+
+void conv1x1_wgrad(torch::Tensor dy, torch::Tensor x, torch::Tensor dw,
+                   int64_t m, int64_t n, int64_t k) {
+  CHECK_IN(dy); CHECK_IN(x); CHECK_IN(dw);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_conv1x1_wgrad(dy.data_ptr<float>(), x.data_ptr<float>(),
+                       dw.data_ptr<float>(), m, (int)n, (int)k, stream);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("conv1x1_fwd", &conv1x1_fwd);
+  mod.def("conv1x1_dgrad", &conv1x1_dgrad);
+  mod.def("conv1x1_wgrad", &conv1x1_wgrad);
   mod.def("ce_topk_fwd", &ce_topk_fwd);
   mod.def("ce_bwd", &ce_bwd);
   mod.def("aug_sample", &aug_sample);

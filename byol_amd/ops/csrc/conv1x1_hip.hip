@@ -1,0 +1,297 @@
+#include "hip/hip_runtime.h"
+// Hand-written f32 MFMA kernels for NHWC 1x1 convolutions (SURVEY.md K1/K2:
+// the 1x1 bottleneck convs are ~half of ResNet-50's conv work; reference
+// reaches them through cuDNN).  gfx950 f32-input MFMA
+// (v_mfma_f32_32x32x2_f32) runs at the full f32 vector rate (155 TF
+// measured) and frees the VALU for addressing/epilogue.
+//
+//   fwd:   Y[M,N]  = X[M,K] . W^T        (W stored [N,K] = torch [Cout,Cin])
+//   dgrad: dX[M,K] = dY[M,N] . W         (W natural [N,K])
+//   wgrad: dW[N,K] = sum_m dY[m,n]*X[m,k]  (M-chunked, fp32 atomics)
+//
+// Structure (correctness-first rung of the guide's GEMM ladder): 128x128
+// block tile, BK=32, 4 waves each owning a 64x64 quadrant as 2x2 MFMA tiles
+// of 32x32, LDS staging with +1-float padding against bank conflicts,
+// single-buffered.  The per-shape dispatch in byol_amd/ops/conv.py enables
+// these only where the microbench (tools/conv_microbench.py) beats MIOpen.
+#include "common.h"
+
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+#define BM 128
+#define BN 128
+#define BK 32
+
+// TRANS_B = false: B stored [Kd][N] row-major (dgrad: W[N,K] with Kd=N, N=K)
+// TRANS_B = true:  B stored [N][Kd] row-major (fwd: W[N,K], Kd=K)
+template <bool TRANS_B>
+__global__ __launch_bounds__(256)
+void conv1x1_gemm_kernel(const float* __restrict__ A,
+                         const float* __restrict__ B,
+                         float* __restrict__ Cmat,
+                         int64_t M, int Kd, int N) {
+  __shared__ float lds_a[BM][BK + 1];
+  __shared__ float lds_b[BK][BN + 1];
+
+  const int ntiles_n = (N + BN - 1) / BN;
+  const int tile_m = blockIdx.x / ntiles_n;
+  const int tile_n = blockIdx.x % ntiles_n;
+  const int64_t m0 = (int64_t)tile_m * BM;
+  const int n0 = tile_n * BN;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = (wave >> 1) * 64;  // wave row offset in tile
+  const int wc = (wave & 1) * 64;   // wave col offset
+
+  f32x16 acc[2][2];
+  #pragma unroll
+  for (int t = 0; t < 2; ++t)
+    #pragma unroll
+    for (int u = 0; u < 2; ++u)
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) acc[t][u][e] = 0.f;
+
+  for (int k0 = 0; k0 < Kd; k0 += BK) {
+    // stage A: BM x BK floats; thread t loads 16 floats of row (t>>1),
+    // half (t&1)
+    {
+      const int r = threadIdx.x >> 1;
+      const int cbase = (threadIdx.x & 1) * 16;
+      const int64_t gr = m0 + r;
+      #pragma unroll
+      for (int c4 = 0; c4 < 4; ++c4) {
+        const int c = cbase + c4 * 4;
+        float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+        if (gr < M && k0 + c + 3 < Kd + 0) {
+          const float* src = A + gr * Kd + k0 + c;
+          v = *reinterpret_cast<const float4*>(src);
+        }
+        lds_a[r][c + 0] = v.x;
+        lds_a[r][c + 1] = v.y;
+        lds_a[r][c + 2] = v.z;
+        lds_a[r][c + 3] = v.w;
+      }
+    }
+    // stage B: BK x BN floats
+    if (!TRANS_B) {
+      // B[k][n]: row k contiguous in n; thread t loads 16 floats of
+      // row (t>>4) at col (t&15)*8? -> BK=32 rows x BN=128: 4096 floats,
+      // 256 threads x 16: thread covers row r=t>>1, 64-col half (t&1)
+      const int r = threadIdx.x >> 3;       // 32 rows
+      const int cbase = (threadIdx.x & 7) * 16;
+      #pragma unroll
+      for (int c4 = 0; c4 < 4; ++c4) {
+        const int c = cbase + c4 * 4;
+        float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+        if (n0 + c + 3 < N) {
+          const float* src = B + (int64_t)(k0 + r) * N + n0 + c;
+          v = *reinterpret_cast<const float4*>(src);
+        } else if (n0 + c < N) {
+          v.x = B[(int64_t)(k0 + r) * N + n0 + c];
+          if (n0 + c + 1 < N) v.y = B[(int64_t)(k0 + r) * N + n0 + c + 1];
+          if (n0 + c + 2 < N) v.z = B[(int64_t)(k0 + r) * N + n0 + c + 2];
+        }
+        lds_b[r][c + 0] = v.x;
+        lds_b[r][c + 1] = v.y;
+        lds_b[r][c + 2] = v.z;
+        lds_b[r][c + 3] = v.w;
+      }
+    } else {
+      // B stored [N][Kd]; we need lds_b[k][n] = B[n0+n][k0+k].
+      // thread t loads 16 floats along Kd of row n = t>>1 (coalesced),
+      // writes transposed.
+      const int n = threadIdx.x >> 1;       // 128 n-rows
+      const int kbase = (threadIdx.x & 1) * 16;
+      #pragma unroll
+      for (int k4 = 0; k4 < 4; ++k4) {
+        const int k = kbase + k4 * 4;
+        float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+        if (n0 + n < N) {
+          const float* src = B + (int64_t)(n0 + n) * Kd + k0 + k;
+          v = *reinterpret_cast<const float4*>(src);
+        }
+        lds_b[k + 0][n] = v.x;
+        lds_b[k + 1][n] = v.y;
+        lds_b[k + 2][n] = v.z;
+        lds_b[k + 3][n] = v.w;
+      }
+    }
+    __syncthreads();
+
+    // MFMA over the staged K-tile: v_mfma_f32_32x32x2_f32, K=2 per instr
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 2) {
+      const int krow = kk + (lane >> 5);     // this lane's k
+      #pragma unroll
+      for (int t = 0; t < 2; ++t) {
+        const float a = lds_a[wr + t * 32 + (lane & 31)][krow];
+        #pragma unroll
+        for (int u = 0; u < 2; ++u) {
+          const float b = lds_b[krow][wc + u * 32 + (lane & 31)];
+          acc[t][u] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc[t][u],
+                                                           0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C/D layout of 32x32 MFMA: col = lane&31,
+  // row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+  #pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        const int row = (e & 3) + 8 * (e >> 2) + 4 * (lane >> 5);
+        const int col = lane & 31;
+        const int64_t gm = m0 + wr + t * 32 + row;
+        const int gn = n0 + wc + u * 32 + col;
+        if (gm < M && gn < N) Cmat[gm * N + gn] = acc[t][u][e];
+      }
+    }
+  }
+}
+
+// wgrad: dW[N,K] += sum over an M-chunk of dY[m,n]*X[m,k].
+// grid.x = n-tiles * k-tiles * m-chunks. A-fragment = dY^T, B = X; both
+// coalesced from global through LDS.
+__global__ __launch_bounds__(256)
+void conv1x1_wgrad_kernel(const float* __restrict__ dY,
+                          const float* __restrict__ X,
+                          float* __restrict__ dW,
+                          int64_t M, int N, int K, int64_t chunk) {
+  __shared__ float lds_dy[BK][BM + 1];  // [m-sub][n]
+  __shared__ float lds_x[BK][BN + 1];   // [m-sub][k]
+
+  const int ntiles_k = (K + BN - 1) / BN;
+  const int ntiles_n = (N + BM - 1) / BM;
+  const int tile_n = (blockIdx.x / ntiles_k) % ntiles_n;
+  const int tile_k = blockIdx.x % ntiles_k;
+  const int64_t mchunk = blockIdx.x / (ntiles_k * ntiles_n);
+  const int n0 = tile_n * BM;
+  const int k0 = tile_k * BN;
+  const int64_t mstart = mchunk * chunk;
+  const int64_t mend = (mstart + chunk < M) ? mstart + chunk : M;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+
+  f32x16 acc[2][2];
+  #pragma unroll
+  for (int t = 0; t < 2; ++t)
+    #pragma unroll
+    for (int u = 0; u < 2; ++u)
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) acc[t][u][e] = 0.f;
+
+  for (int64_t m0 = mstart; m0 < mend; m0 += BK) {
+    // stage dY rows [m0..m0+32) x n-tile 128 -> lds_dy[m][n]
+    {
+      const int r = threadIdx.x >> 3;           // 32 m-rows
+      const int cbase = (threadIdx.x & 7) * 16;
+      #pragma unroll
+      for (int c4 = 0; c4 < 4; ++c4) {
+        const int c = cbase + c4 * 4;
+        float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+        if (m0 + r < mend && n0 + c + 3 < N) {
+          v = *reinterpret_cast<const float4*>(
+              dY + (m0 + r) * N + n0 + c);
+        } else if (m0 + r < mend) {
+          if (n0 + c < N) v.x = dY[(m0 + r) * N + n0 + c];
+          if (n0 + c + 1 < N) v.y = dY[(m0 + r) * N + n0 + c + 1];
+          if (n0 + c + 2 < N) v.z = dY[(m0 + r) * N + n0 + c + 2];
+        }
+        lds_dy[r][c + 0] = v.x;
+        lds_dy[r][c + 1] = v.y;
+        lds_dy[r][c + 2] = v.z;
+        lds_dy[r][c + 3] = v.w;
+      }
+    }
+    {
+      const int r = threadIdx.x >> 3;
+      const int cbase = (threadIdx.x & 7) * 16;
+      #pragma unroll
+      for (int c4 = 0; c4 < 4; ++c4) {
+        const int c = cbase + c4 * 4;
+        float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+        if (m0 + r < mend && k0 + c + 3 < K) {
+          v = *reinterpret_cast<const float4*>(X + (m0 + r) * K + k0 + c);
+        } else if (m0 + r < mend) {
+          if (k0 + c < K) v.x = X[(m0 + r) * K + k0 + c];
+          if (k0 + c + 1 < K) v.y = X[(m0 + r) * K + k0 + c + 1];
+          if (k0 + c + 2 < K) v.z = X[(m0 + r) * K + k0 + c + 2];
+        }
+        lds_x[r][c + 0] = v.x;
+        lds_x[r][c + 1] = v.y;
+        lds_x[r][c + 2] = v.z;
+        lds_x[r][c + 3] = v.w;
+      }
+    }
+    __syncthreads();
+
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 2) {
+      const int mrow = kk + (lane >> 5);
+      #pragma unroll
+      for (int t = 0; t < 2; ++t) {
+        const float a = lds_dy[mrow][wr + t * 32 + (lane & 31)];
+        #pragma unroll
+        for (int u = 0; u < 2; ++u) {
+          const float b = lds_x[mrow][wc + u * 32 + (lane & 31)];
+          acc[t][u] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc[t][u],
+                                                           0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  #pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        const int row = (e & 3) + 8 * (e >> 2) + 4 * (lane >> 5);
+        const int col = lane & 31;
+        const int gn = n0 + wr + t * 32 + row;   // output row = n
+        const int gk = k0 + wc + u * 32 + col;   // output col = k
+        if (gn < N && gk < K)
+          atomicAdd(&dW[(int64_t)gn * K + gk], acc[t][u][e]);
+      }
+    }
+  }
+}
+
+static inline int64_t cdiv(int64_t a, int64_t b) { return (a + b - 1) / b; }
+
+void launch_conv1x1_fwd(const float* x, const float* w, float* y, int64_t m,
+                        int k, int n, hipStream_t stream) {
+  const int64_t grid = cdiv(m, BM) * cdiv(n, BN);
+  hipLaunchKernelGGL((conv1x1_gemm_kernel<true>), dim3((uint32_t)grid),
+                     dim3(256), 0, stream, x, w, y, m, k, n);
+}
+
+void launch_conv1x1_dgrad(const float* dy, const float* w, float* dx,
+                          int64_t m, int n, int k, hipStream_t stream) {
+  // dX[M,K] = dY[M,N] . W[N,K]  (A=dY, Kd=N, output N-dim = K)
+  const int64_t grid = cdiv(m, BM) * cdiv(k, BN);
+  hipLaunchKernelGGL((conv1x1_gemm_kernel<false>), dim3((uint32_t)grid),
+                     dim3(256), 0, stream, dy, w, dx, m, n, k);
+}
+
+void launch_conv1x1_wgrad(const float* dy, const float* x, float* dw,
+                          int64_t m, int n, int k, hipStream_t stream) {
+  // chunk M so total blocks ~<= 4096 per (n,k) tile-grid
+  const int64_t tiles = cdiv(n, BM) * cdiv(k, BN);
+  int64_t chunk = 32 * BK;  // 1024 rows minimum
+  while (cdiv(m, chunk) * tiles > 4096) chunk *= 2;
+  const int64_t grid = tiles * cdiv(m, chunk);
+  hipLaunchKernelGGL(conv1x1_wgrad_kernel, dim3((uint32_t)grid), dim3(256),
+                     0, stream, dy, x, dw, m, n, k, chunk);
+}

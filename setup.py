@@ -22,6 +22,7 @@ sources = [
     os.path.join(CSRC, "bn_fused.hip"),
     os.path.join(CSRC, "augment.hip"),
     os.path.join(CSRC, "cetopk.hip"),
+    os.path.join(CSRC, "conv1x1.hip"),
 ]
 
 setup(

@@ -1,0 +1,95 @@
+#!/usr/bin/env python3
+"""Per-shape microbench: MFMA conv1x1 kernels vs MIOpen (F.conv2d), fwd and
+bwd, on the ResNet-50 bs=512 1x1 shapes.  Also checks numerics vs MIOpen.
+Run on an MI355X:  python tools/conv_microbench.py [--batch 512]
+"""
+
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+os.environ.setdefault("MIOPEN_FIND_MODE", "NORMAL")
+os.environ["BYOL_MFMA_CONV1X1"] = "1"
+
+import torch
+import torch.nn.functional as F
+
+# (H, Cin, Cout) stride-1 1x1 shapes in ResNet-50 @224
+SHAPES = [
+    (56, 64, 64), (56, 64, 256), (56, 256, 64), (56, 256, 128),
+    (28, 128, 512), (28, 512, 128), (28, 512, 256),
+    (14, 256, 1024), (14, 1024, 256), (14, 1024, 512),
+    (7, 512, 2048), (7, 2048, 512),
+]
+
+
+def timeit(fn, iters=10, warmup=3):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1000
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--batch", type=int, default=512)
+    ap.add_argument("--iters", type=int, default=10)
+    args = ap.parse_args()
+    from byol_amd.ops import require_extension
+    from byol_amd.ops.conv import _Conv1x1Fn
+    require_extension("microbench")
+    torch.manual_seed(0)
+    print(f"{'shape':>22} {'miopen f':>9} {'mfma f':>9} {'miopen b':>9} "
+          f"{'mfma b':>9}  maxerr_f maxerr_dx maxerr_dw")
+    for hgt, cin, cout in SHAPES:
+        b = args.batch
+        x = torch.randn(b, cin, hgt, hgt, device="cuda").to(
+            memory_format=torch.channels_last)
+        w = torch.randn(cout, cin, 1, 1, device="cuda") * 0.05
+
+        # numerics
+        x1 = x.clone().requires_grad_(True)
+        w1 = w.clone().requires_grad_(True)
+        y1 = F.conv2d(x1, w1)
+        g = torch.randn_like(y1).to(memory_format=torch.channels_last)
+        y1.backward(g)
+        x2 = x.clone().requires_grad_(True)
+        w2 = w.clone().requires_grad_(True)
+        y2 = _Conv1x1Fn.apply(x2, w2)
+        y2.backward(g)
+        ef = (y1 - y2).abs().max().item()
+        edx = (x1.grad - x2.grad).abs().max().item()
+        edw = (w1.grad - w2.grad).abs().max().item() / max(
+            w1.grad.abs().max().item(), 1e-6)
+
+        # perf: forward
+        tm_f = timeit(lambda: F.conv2d(x, w), args.iters)
+        to_f = timeit(lambda: _Conv1x1Fn.apply(x, w), args.iters)
+
+        # perf: full fwd+bwd
+        def bwd_miopen():
+            xr = x.detach().requires_grad_(True)
+            wr = w.detach().requires_grad_(True)
+            F.conv2d(xr, wr).backward(g)
+
+        def bwd_ours():
+            xr = x.detach().requires_grad_(True)
+            wr = w.detach().requires_grad_(True)
+            _Conv1x1Fn.apply(xr, wr).backward(g)
+
+        tm_b = timeit(bwd_miopen, args.iters)
+        to_b = timeit(bwd_ours, args.iters)
+        tag = "<<" if to_f < tm_f and to_b < tm_b else ""
+        print(f"H{hgt:>3} K{cin:>5} N{cout:>5} "
+              f"{tm_f:9.3f} {to_f:9.3f} {tm_b:9.3f} {to_b:9.3f}  "
+              f"{ef:.2e} {edx:.2e} {edw:.2e} {tag}")
+
+
+if __name__ == "__main__":
+    main()
