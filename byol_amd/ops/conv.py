@@ -61,9 +61,11 @@ class _Conv1x1Fn(torch.autograd.Function):
         n = weight.shape[0]
         m = b * h * w
         wv = weight.reshape(n, k).contiguous()
+        # pre-transposed copy feeds the glds fast path ([K,N] natural)
+        wt = wv.t().contiguous()
         y = torch.empty(b, n, h, w, device=x.device, dtype=x.dtype) \
             .to(memory_format=torch.channels_last)
-        C.conv1x1_fwd(_rows(x, k), wv, _rows(y, n), m, k, n)
+        C.conv1x1_fwd(_rows(x, k), wv, wt, _rows(y, n), m, k, n)
         ctx.save_for_backward(x, wv)
         ctx.dims = (b, k, h, w, n, m)
         return y

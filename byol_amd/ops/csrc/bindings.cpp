@@ -47,8 +47,9 @@ void launch_ce_topk_fwd(const float* logits, const int64_t* labels,
 void launch_ce_bwd(const float* logits, const int64_t* labels,
                    const float* row_stats, const float* grad_out,
                    float* dlogits, int m, int n, hipStream_t stream);
-void launch_conv1x1_fwd(const float* x, const float* w, float* y, int64_t m,
-                        int k, int n, hipStream_t stream);
+void launch_conv1x1_fwd(const float* x, const float* w, const float* wt,
+                        float* y, int64_t m, int k, int n,
+                        hipStream_t stream);
 void launch_conv1x1_dgrad(const float* dy, const float* w, float* dx,
                           int64_t m, int n, int k, hipStream_t stream);
 void launch_conv1x1_wgrad(const float* dy, const float* x, float* dw,
@@ -259,11 +260,13 @@ void ce_bwd(torch::Tensor logits, torch::Tensor labels,
                 (int)logits.size(1), stream);
 }
 
-void conv1x1_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor y,
+void conv1x1_fwd(torch::Tensor x, torch::Tensor w,
+                 c10::optional<torch::Tensor> wt, torch::Tensor y,
                  int64_t m, int64_t k, int64_t n) {
   CHECK_IN(x); CHECK_IN(w); CHECK_IN(y);
   auto stream = at::hip::getCurrentHIPStream();
   launch_conv1x1_fwd(x.data_ptr<float>(), w.data_ptr<float>(),
+                     wt ? wt->data_ptr<float>() : nullptr,
                      y.data_ptr<float>(), m, (int)k, (int)n, stream);
 }
 

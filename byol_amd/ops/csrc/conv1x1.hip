@@ -267,21 +267,146 @@ void conv1x1_wgrad_kernel(const float* __restrict__ dY,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fast path: glds (global_load_lds) double-buffered 2-phase pipeline for
+// full tiles (M%128==0, N%128==0, K%32==0).  B must be [Kd][N] natural
+// (the host pre-transposes W for fwd).  A-tile LDS image is quad-rotated
+// (source-side swizzle, guide rule 21) so the column-wise A-fragment reads
+// are 4-way instead of 32-way bank conflicted.
+// ---------------------------------------------------------------------------
+#define LDSW_A 4096   // 128x32 floats
+#define LDSW_B 4096   // 32x128 floats
+#define LDSW_BUF (LDSW_A + LDSW_B)
+
+__global__ __launch_bounds__(256)
+void conv1x1_gemm_fast_kernel(const float* __restrict__ A,
+                              const float* __restrict__ B,
+                              float* __restrict__ Cmat,
+                              int64_t M, int Kd, int N) {
+  __shared__ __attribute__((aligned(16))) float lds[2 * LDSW_BUF];
+
+  const int ntiles_n = N / BN;
+  const int tile_m = blockIdx.x / ntiles_n;
+  const int tile_n = blockIdx.x % ntiles_n;
+  const int64_t m0 = (int64_t)tile_m * BM;
+  const int n0 = tile_n * BN;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+
+  f32x16 acc[2][2];
+  #pragma unroll
+  for (int t = 0; t < 2; ++t)
+    #pragma unroll
+    for (int u = 0; u < 2; ++u)
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) acc[t][u][e] = 0.f;
+
+  // per-lane source geometry for the 4 A-issues and 4 B-issues of this wave
+  // A dest word = (wave*4+i)*256 + lane*4 -> row r=off>>5, quad q=(off&31)>>2
+  // source quad rotated: g = (q - r) & 7
+  auto stage = [&](int buf, int k0) {
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int off = (wave * 4 + i) * 256 + lane * 4;
+      const int r = off >> 5;
+      const int q = (off & 31) >> 2;
+      const float* src = A + (m0 + r) * (int64_t)Kd + k0 + (((q - r) & 7) << 2);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * LDSW_BUF + (wave * 4 + i) * 256],
+          16, 0, 0);
+    }
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int off = (wave * 4 + i) * 256 + lane * 4;
+      const int r = off >> 7;            // 32 k-rows
+      const int c = off & 127;           // 128 n-cols
+      const float* src = B + (int64_t)(k0 + r) * N + n0 + c;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * LDSW_BUF + LDSW_A + (wave * 4 + i) * 256],
+          16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  const int ntiles_k = Kd / BK;
+  int buf = 0;
+  for (int kt = 0; kt < ntiles_k; ++kt) {
+    if (kt + 1 < ntiles_k) stage(buf ^ 1, (kt + 1) * BK);
+    const float* la = &lds[buf * LDSW_BUF];
+    const float* lb = &lds[buf * LDSW_BUF + LDSW_A];
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 2) {
+      const int krow = kk + (lane >> 5);
+      #pragma unroll
+      for (int t = 0; t < 2; ++t) {
+        const int row = wr + t * 32 + (lane & 31);
+        // swizzled A image: element k of row r lives at quad (k/4+r)&7
+        const float a = la[row * 32 + ((((krow >> 2) + row) & 7) << 2)
+                           + (krow & 3)];
+        #pragma unroll
+        for (int u = 0; u < 2; ++u) {
+          const float b = lb[krow * BN + wc + u * 32 + (lane & 31)];
+          acc[t][u] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc[t][u],
+                                                           0, 0, 0);
+        }
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+
+  #pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    #pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        const int row = (e & 3) + 8 * (e >> 2) + 4 * (lane >> 5);
+        const int col = lane & 31;
+        Cmat[(m0 + wr + t * 32 + row) * N + n0 + wc + u * 32 + col] =
+            acc[t][u][e];
+      }
+    }
+  }
+}
+
 static inline int64_t cdiv(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
-void launch_conv1x1_fwd(const float* x, const float* w, float* y, int64_t m,
-                        int k, int n, hipStream_t stream) {
+void launch_conv1x1_fwd(const float* x, const float* w, const float* wt,
+                        float* y, int64_t m, int k, int n,
+                        hipStream_t stream) {
   const int64_t grid = cdiv(m, BM) * cdiv(n, BN);
-  hipLaunchKernelGGL((conv1x1_gemm_kernel<true>), dim3((uint32_t)grid),
-                     dim3(256), 0, stream, x, w, y, m, k, n);
+  if (m % BM == 0 && n % BN == 0 && k % BK == 0 && wt != nullptr) {
+    hipLaunchKernelGGL(conv1x1_gemm_fast_kernel, dim3((uint32_t)grid),
+                       dim3(256), 0, stream, x, wt, y, m, k, n);
+  } else {
+    hipLaunchKernelGGL((conv1x1_gemm_kernel<true>), dim3((uint32_t)grid),
+                       dim3(256), 0, stream, x, w, y, m, k, n);
+  }
 }
 
 void launch_conv1x1_dgrad(const float* dy, const float* w, float* dx,
                           int64_t m, int n, int k, hipStream_t stream) {
   // dX[M,K] = dY[M,N] . W[N,K]  (A=dY, Kd=N, output N-dim = K)
   const int64_t grid = cdiv(m, BM) * cdiv(k, BN);
-  hipLaunchKernelGGL((conv1x1_gemm_kernel<false>), dim3((uint32_t)grid),
-                     dim3(256), 0, stream, dy, w, dx, m, n, k);
+  if (m % BM == 0 && k % BN == 0 && n % BK == 0) {
+    hipLaunchKernelGGL(conv1x1_gemm_fast_kernel, dim3((uint32_t)grid),
+                       dim3(256), 0, stream, dy, w, dx, m, n, k);
+  } else {
+    hipLaunchKernelGGL((conv1x1_gemm_kernel<false>), dim3((uint32_t)grid),
+                       dim3(256), 0, stream, dy, w, dx, m, n, k);
+  }
 }
 
 void launch_conv1x1_wgrad(const float* dy, const float* x, float* dw,
