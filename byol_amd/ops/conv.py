@@ -27,10 +27,14 @@ def mfma_conv_mode() -> str:
     return os.environ.get("BYOL_MFMA_CONV1X1", "auto")
 
 
-# Shapes (K=Cin, N=Cout) where the MFMA kernel measured >= MIOpen on
-# MI355X at bs-512-class M (see profiles/ microbench results); "auto" mode
-# enables exactly these.  Updated from tools/conv_microbench.py runs.
-_AUTO_SHAPES = set()
+# Shapes (K=Cin, N=Cout) where the MFMA FORWARD kernel measured faster
+# than MIOpen on MI355X at bs-512-class M (tools/conv_microbench.py,
+# profiles/r01_conv1x1_microbench.md); "auto" mode enables exactly these
+# and routes backward to MIOpen (aten.convolution_backward), which still
+# wins there.  BYOL_MFMA_CONV1X1=1 forces our kernels everywhere (tests).
+_AUTO_SHAPES = {
+    (64, 256), (256, 128), (128, 512), (512, 128), (512, 256), (256, 1024),
+}
 
 
 def _eligible(x: torch.Tensor, weight: torch.Tensor) -> bool:
@@ -63,8 +67,8 @@ class _Conv1x1Fn(torch.autograd.Function):
         wv = weight.reshape(n, k).contiguous()
         # pre-transposed copy feeds the glds fast path ([K,N] natural)
         wt = wv.t().contiguous()
-        y = torch.empty(b, n, h, w, device=x.device, dtype=x.dtype) \
-            .to(memory_format=torch.channels_last)
+        y = torch.empty((b, n, h, w), device=x.device, dtype=x.dtype,
+                        memory_format=torch.channels_last)
         C.conv1x1_fwd(_rows(x, k), wv, wt, _rows(y, n), m, k, n)
         ctx.save_for_backward(x, wv)
         ctx.dims = (b, k, h, w, n, m)
@@ -72,15 +76,23 @@ class _Conv1x1Fn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        C = extension()
         x, wv = ctx.saved_tensors
         b, k, h, w, n, m = ctx.dims
         dy = dy.contiguous(memory_format=torch.channels_last)
-        dx = torch.empty_like(x)
-        C.conv1x1_dgrad(_rows(dy, n), wv, _rows(dx, k), m, n, k)
-        dw = torch.zeros(n, k, device=x.device, dtype=x.dtype)
-        C.conv1x1_wgrad(_rows(dy, n), _rows(x, k), dw, m, n, k)
-        return dx, dw.view(n, k, 1, 1)
+        if mfma_conv_mode() == "1":
+            # our dgrad/wgrad kernels (forced mode; numerics tests)
+            C = extension()
+            dx = torch.empty_like(x)
+            C.conv1x1_dgrad(_rows(dy, n), wv, _rows(dx, k), m, n, k)
+            dw = torch.zeros(n, k, device=x.device, dtype=x.dtype)
+            C.conv1x1_wgrad(_rows(dy, n), _rows(x, k), dw, m, n, k)
+            return dx, dw.view(n, k, 1, 1)
+        # auto mode: MIOpen backward still wins these shapes (see
+        # profiles/r01_conv1x1_microbench.md) — route dgrad/wgrad there
+        dx, dw, _ = torch.ops.aten.convolution_backward(
+            dy, x, wv.view(n, k, 1, 1), [0], [1, 1], [0, 0], [1, 1], False,
+            [0, 0], 1, [True, True, False])
+        return dx, dw
 
 
 class MFMAConv1x1(nn.Conv2d):

@@ -366,18 +366,34 @@ void conv1x1_gemm_fast_kernel(const float* __restrict__ A,
     buf ^= 1;
   }
 
+  // Epilogue via LDS re-stage: the raw MFMA C-layout would need 64
+  // scattered dword stores per lane (store-ISSUE-bound — the dominant cost
+  // for small-K shapes).  Each wave transposes its 64x64 quadrant through
+  // its own 16 KB LDS slice and stores contiguous dwordx4 rows instead.
+  __syncthreads();  // everyone done reading the staging buffers
+  float* cw = &lds[wave * 4096];
   #pragma unroll
   for (int t = 0; t < 2; ++t) {
     #pragma unroll
     for (int u = 0; u < 2; ++u) {
       #pragma unroll
       for (int e = 0; e < 16; ++e) {
-        const int row = (e & 3) + 8 * (e >> 2) + 4 * (lane >> 5);
-        const int col = lane & 31;
-        Cmat[(m0 + wr + t * 32 + row) * N + n0 + wc + u * 32 + col] =
-            acc[t][u][e];
+        const int lrow = t * 32 + (e & 3) + 8 * (e >> 2) + 4 * (lane >> 5);
+        const int lcol = u * 32 + (lane & 31);
+        cw[lrow * 64 + lcol] = acc[t][u][e];
       }
     }
+  }
+  __builtin_amdgcn_s_barrier();  // intra-wave only needed; cheap
+  const int64_t gm_base = m0 + wr;
+  const int gn_base = n0 + wc;
+  #pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    const int lrow = p * 4 + (lane >> 4);
+    const int lcol = (lane & 15) * 4;
+    const float4 v = *reinterpret_cast<const float4*>(&cw[lrow * 64 + lcol]);
+    *reinterpret_cast<float4*>(
+        &Cmat[(gm_base + lrow) * N + gn_base + lcol]) = v;
   }
 }
 

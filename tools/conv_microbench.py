@@ -68,9 +68,17 @@ def main():
         edw = (w1.grad - w2.grad).abs().max().item() / max(
             w1.grad.abs().max().item(), 1e-6)
 
-        # perf: forward
+        # perf: forward (wrapper) + raw kernel (no alloc/transpose)
         tm_f = timeit(lambda: F.conv2d(x, w), args.iters)
         to_f = timeit(lambda: _Conv1x1Fn.apply(x, w), args.iters)
+        C = require_extension("raw")
+        m = b * hgt * hgt
+        xr = x.permute(0, 2, 3, 1).reshape(-1, cin)
+        wv = w.reshape(cout, cin).contiguous()
+        wt = wv.t().contiguous()
+        yr = torch.empty(m, cout, device="cuda")
+        t_raw = timeit(lambda: C.conv1x1_fwd(xr, wv, wt, yr, m, cin, cout),
+                       args.iters)
 
         # perf: full fwd+bwd
         def bwd_miopen():
@@ -87,8 +95,8 @@ def main():
         to_b = timeit(bwd_ours, args.iters)
         tag = "<<" if to_f < tm_f and to_b < tm_b else ""
         print(f"H{hgt:>3} K{cin:>5} N{cout:>5} "
-              f"{tm_f:9.3f} {to_f:9.3f} {tm_b:9.3f} {to_b:9.3f}  "
-              f"{ef:.2e} {edx:.2e} {edw:.2e} {tag}")
+              f"{tm_f:9.3f} {to_f:9.3f} raw{t_raw:8.3f} {tm_b:9.3f} "
+              f"{to_b:9.3f}  {ef:.2e} {edx:.2e} {edw:.2e} {tag}")
 
 
 if __name__ == "__main__":
