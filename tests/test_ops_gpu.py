@@ -329,6 +329,32 @@ def test_gpu_two_view_pipeline_end_to_end(ext):
     assert not torch.equal(a1, a2)
 
 
+def test_mfma_conv1x1_vs_miopen(ext, monkeypatch):
+    """Forced MFMA mode: fwd/dgrad/wgrad kernels vs F.conv2d oracle."""
+    import torch.nn.functional as F
+    from byol_amd.ops.conv import _Conv1x1Fn
+    monkeypatch.setenv("BYOL_MFMA_CONV1X1", "1")
+    torch.manual_seed(12)
+    for bsz, hgt, cin, cout in [(8, 56, 64, 256), (8, 28, 512, 128),
+                                (4, 7, 2048, 512), (4, 14, 96, 224)]:
+        x = torch.randn(bsz, cin, hgt, hgt, device="cuda").to(
+            memory_format=torch.channels_last).requires_grad_(True)
+        w = (torch.randn(cout, cin, 1, 1, device="cuda") * 0.05
+             ).requires_grad_(True)
+        y = _Conv1x1Fn.apply(x, w)
+        g = torch.randn_like(y).to(memory_format=torch.channels_last)
+        y.backward(g)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        y2 = F.conv2d(x2, w2)
+        y2.backward(g)
+        assert torch.allclose(y, y2, rtol=1e-4, atol=1e-4), \
+            (hgt, cin, cout, (y - y2).abs().max().item())
+        assert torch.allclose(x.grad, x2.grad, rtol=1e-4, atol=1e-4)
+        assert torch.allclose(w.grad, w2.grad, rtol=1e-3, atol=1e-2), \
+            (w.grad - w2.grad).abs().max().item()
+
+
 def test_smoke_entrypoint():
     import __graft_entry__
     __graft_entry__.smoke()
