@@ -351,9 +351,14 @@ def run(rank, args):
             break
 
         if epoch == 2 and args.distributed_rank == 0 and grapher is not None:
+            from ..utils import get_slurm_id
+            config_to_post = dict(vars(args))
+            slurm_id = get_slurm_id()
+            if slurm_id is not None:
+                config_to_post["slurm_job_id"] = slurm_id
             grapher.add_text(
                 "config",
-                pprint.PrettyPrinter(indent=4).pformat(vars(args)), 0)
+                pprint.PrettyPrinter(indent=4).pformat(config_to_post), 0)
 
     if grapher is not None:
         grapher.close()

@@ -41,7 +41,8 @@ __all__ = ["FlatDDP"]
 
 
 class _Bucket:
-    __slots__ = ("start", "end", "param_ids", "pending", "work", "launched")
+    __slots__ = ("start", "end", "param_ids", "pending", "work", "launched",
+                 "ready")
 
     def __init__(self, start: int, end: int):
         self.start = start
@@ -50,6 +51,7 @@ class _Bucket:
         self.pending = 0
         self.work = None
         self.launched = False
+        self.ready = False
 
 
 class FlatDDP(nn.Module):
@@ -103,6 +105,8 @@ class FlatDDP(nn.Module):
             b.pending = len(b.param_ids)
             b.work = None
             b.launched = False
+            b.ready = False
+        self._next_launch = 0
 
     def _make_hook(self, pid: int):
         def hook(_param):
@@ -110,9 +114,21 @@ class FlatDDP(nn.Module):
                 return
             b = self._param_bucket[pid]
             b.pending -= 1
-            if b.pending == 0 and not b.launched:
-                self._launch(b)
+            if b.pending == 0:
+                b.ready = True
+                self._drain_ready()
         return hook
+
+    def _drain_ready(self):
+        """Launch ready buckets strictly in bucket order: RCCL requires the
+        same collective order on every rank, and hook completion order is
+        not architecturally guaranteed to match across ranks."""
+        while self._next_launch < len(self.buckets):
+            b = self.buckets[self._next_launch]
+            if not getattr(b, "ready", False):
+                break
+            self._launch(b)
+            self._next_launch += 1
 
     def _launch(self, b: _Bucket):
         view = self._space.flat_grads[b.start:b.end]
@@ -126,11 +142,14 @@ class FlatDDP(nn.Module):
     require_backward_grad_sync = True
 
     def finish_grad_sync(self):
-        """Launch any not-yet-fired buckets and wait for all; call after
-        backward, before the optimizer step."""
+        """Launch any not-yet-fired buckets (in order) and wait for all;
+        call after backward, before the optimizer step.  Buckets whose
+        params got no grad this step all-reduce zeros — the
+        find_unused_parameters semantics without the graph walk."""
         for b in self.buckets:
             if not b.launched:
                 self._launch(b)
+        self._next_launch = len(self.buckets)
         for b in self.buckets:
             if b.work is not None:
                 b.work.wait()

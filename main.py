@@ -10,6 +10,10 @@ from byol_amd.parallel import launch
 
 def main():
     args = parse_args()
+    from byol_amd.utils import get_aws_instance_id
+    instance_id = get_aws_instance_id()
+    if instance_id is not None:
+        args.instance_id = instance_id
     launch(run, args)
 
 
