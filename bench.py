@@ -64,6 +64,10 @@ def parse():
     p.add_argument("--bucket-cap-mb", type=float, default=32.0)
     p.add_argument("--miopen-benchmark", action="store_true", default=False,
                    help="exhaustive MIOpen find (slow first run)")
+    p.add_argument("--debug-cpu", action="store_true", default=False,
+                   help="plumbing mode: tiny model on CPU over gloo "
+                        "(validates the torchrun/distributed path without "
+                        "GPUs; not a performance measurement)")
     return p.parse_args()
 
 
@@ -72,12 +76,20 @@ def main():
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
-    assert torch.cuda.is_available(), "bench.py needs a GPU"
-    torch.cuda.set_device(local_rank)
-    device = torch.device("cuda", local_rank)
+    if args.debug_cpu:
+        device = torch.device("cpu")
+        args.arch = "resnet18"
+        args.batch_per_gpu = min(args.batch_per_gpu, 4)
+        args.image_size = min(args.image_size, 32)
+        args.channels_last = False
+    else:
+        assert torch.cuda.is_available(), "bench.py needs a GPU"
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
     distributed = world > 1
     if distributed:
-        dist.init_process_group(backend="nccl")
+        dist.init_process_group(
+            backend="gloo" if args.debug_cpu else "nccl")
 
     from byol_amd import layers
     from byol_amd.models.byol import BYOL
@@ -118,7 +130,7 @@ def main():
     # augmentation happens upstream of the timed step in real training too
     # when the GPU pipeline is used)
     n_pool = 2
-    gen = torch.Generator(device="cuda").manual_seed(99 + rank)
+    gen = torch.Generator(device=device).manual_seed(99 + rank)
     pool = [(torch.rand(B, 3, S, S, device=device, generator=gen),
              torch.rand(B, 3, S, S, device=device, generator=gen),
              torch.randint(args.classes, (B,), device=device, generator=gen))
@@ -128,7 +140,7 @@ def main():
                  b.to(memory_format=torch.channels_last), l)
                 for a, b, l in pool]
 
-    use_bf16 = args.dtype == "bf16"
+    use_bf16 = args.dtype == "bf16" and device.type == "cuda"
 
     def step(i):
         a1, a2, lab = pool[i % n_pool]
@@ -159,11 +171,13 @@ def main():
 
     if distributed:
         dist.barrier()
-    torch.cuda.synchronize()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for i in range(args.steps):
         step(i)
-    torch.cuda.synchronize()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
     if distributed:
         dist.barrier()
     elapsed = time.perf_counter() - t0
