@@ -7,7 +7,6 @@ fused loss/EMA/LARS HIP kernels)."""
 import functools
 import pprint
 import time
-from typing import Optional
 
 import torch
 import torch.nn as nn
@@ -19,7 +18,6 @@ from ..models.byol import BYOL
 from ..objective import loss_function
 from ..optim import build_optimizer
 from ..parallel import FlatDDP, convert_sync_batchnorm
-from . import metrics
 from .grapher import Grapher
 from .saver import CheckpointBundle, ModelSaver, get_name
 
@@ -150,7 +148,6 @@ def make_grid(images: torch.Tensor, nrow: int = 8,
 def register_plots(loss, grapher, epoch, args, prefix="train"):
     if args.distributed_rank != 0 or grapher is None:
         return
-    import numpy as np
     for k, v in loss.items():
         if isinstance(v, dict):
             register_plots(v, grapher, epoch, args, prefix=prefix)
