@@ -25,6 +25,11 @@ __all__ = ["ResNetEncoder", "build_encoder", "ARCH_SPECS", "arch_names"]
 
 
 def conv3x3(in_planes: int, out_planes: int, stride: int = 1) -> nn.Conv2d:
+    if in_planes % 32 == 0:
+        # MFMA implicit-GEMM forward path (env-gated; MIOpen fallback)
+        from ..ops.conv import MFMAConv3x3
+        return MFMAConv3x3(in_planes, out_planes, kernel_size=3,
+                           stride=stride, padding=1, bias=False)
     return nn.Conv2d(in_planes, out_planes, kernel_size=3, stride=stride,
                      padding=1, bias=False)
 

@@ -104,3 +104,56 @@ class MFMAConv1x1(nn.Conv2d):
             return _Conv1x1Fn.apply(x, self.weight)
         return F.conv2d(x, self.weight, self.bias, self.stride,
                         self.padding, self.dilation, self.groups)
+
+
+# ---------------------------------------------------------------------------
+# 3x3 convolution: MFMA forward (implicit GEMM over 9 taps), MIOpen backward
+# ---------------------------------------------------------------------------
+
+def mfma_conv3x3_mode() -> str:
+    # off by default until the microbench proves per-shape wins (round-2
+    # tuning track); "1" enables everywhere.
+    return os.environ.get("BYOL_MFMA_CONV3X3", "0")
+
+
+class _Conv3x3Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, stride):
+        C = extension()
+        b, k, hi, wi = x.shape
+        n = weight.shape[0]
+        ho = (hi + 2 - 3) // stride + 1
+        wo = (wi + 2 - 3) // stride + 1
+        # prepermute W [N,K,3,3] -> [9][K][N] for coalesced B staging
+        wp = weight.reshape(n, k, 9).permute(2, 1, 0).contiguous()
+        y = torch.empty((b, n, ho, wo), device=x.device, dtype=x.dtype,
+                        memory_format=torch.channels_last)
+        C.conv3x3_fwd(_rows(x, k), wp, _rows(y, n), b, hi, wi, ho, wo, k, n,
+                      stride)
+        ctx.save_for_backward(x, weight)
+        ctx.stride = stride
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        s = ctx.stride
+        dx, dw, _ = torch.ops.aten.convolution_backward(
+            dy, x, weight, [0], [s, s], [1, 1], [1, 1], False, [0, 0], 1,
+            [True, True, False])
+        return dx, dw, None
+
+
+class MFMAConv3x3(nn.Conv2d):
+    """Drop-in for bias-free pad-1 3x3 nn.Conv2d; MFMA forward when enabled
+    and eligible, MIOpen otherwise (and always for backward)."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if (mfma_conv3x3_mode() == "1" and x.is_cuda
+                and x.dtype == torch.float32 and has_extension()
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and self.weight.shape[1] % 32 == 0):
+            return _Conv3x3Fn.apply(x, self.weight, self.stride[0])
+        return F.conv2d(x, self.weight, self.bias, self.stride,
+                        self.padding, self.dilation, self.groups)

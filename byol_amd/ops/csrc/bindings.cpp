@@ -54,6 +54,9 @@ void launch_conv1x1_dgrad(const float* dy, const float* w, float* dx,
                           int64_t m, int n, int k, hipStream_t stream);
 void launch_conv1x1_wgrad(const float* dy, const float* x, float* dw,
                           int64_t m, int n, int k, hipStream_t stream);
+void launch_conv3x3_fwd(const float* x, const float* wp, float* y, int b,
+                        int hi, int wi, int ho, int wo, int k, int n,
+                        int stride, hipStream_t stream);
 void launch_lars_momentum_step(float* p, const float* g, float* m,
                                float* norm_acc, float* alr,
                                const int64_t* seg_off,
@@ -286,9 +289,20 @@ void conv1x1_wgrad(torch::Tensor dy, torch::Tensor x, torch::Tensor dw,
                        dw.data_ptr<float>(), m, (int)n, (int)k, stream);
 }
 
+void conv3x3_fwd(torch::Tensor x, torch::Tensor wp, torch::Tensor y,
+                 int64_t b, int64_t hi, int64_t wi, int64_t ho, int64_t wo,
+                 int64_t k, int64_t n, int64_t stride) {
+  CHECK_IN(x); CHECK_IN(wp); CHECK_IN(y);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_conv3x3_fwd(x.data_ptr<float>(), wp.data_ptr<float>(),
+                     y.data_ptr<float>(), (int)b, (int)hi, (int)wi, (int)ho,
+                     (int)wo, (int)k, (int)n, (int)stride, stream);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("conv3x3_fwd", &conv3x3_fwd);
   mod.def("conv1x1_fwd", &conv1x1_fwd);
   mod.def("conv1x1_dgrad", &conv1x1_dgrad);
   mod.def("conv1x1_wgrad", &conv1x1_wgrad);

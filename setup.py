@@ -23,6 +23,7 @@ sources = [
     os.path.join(CSRC, "augment.hip"),
     os.path.join(CSRC, "cetopk.hip"),
     os.path.join(CSRC, "conv1x1.hip"),
+    os.path.join(CSRC, "conv3x3.hip"),
 ]
 
 setup(

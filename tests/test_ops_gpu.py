@@ -355,6 +355,32 @@ def test_mfma_conv1x1_vs_miopen(ext, monkeypatch):
             (w.grad - w2.grad).abs().max().item()
 
 
+def test_mfma_conv3x3_vs_miopen(ext, monkeypatch):
+    import torch.nn.functional as F
+    from byol_amd.ops.conv import _Conv3x3Fn
+    torch.manual_seed(13)
+    for bsz, hgt, cin, cout, stride in [(4, 14, 64, 64, 1),
+                                        (4, 15, 128, 128, 1),
+                                        (4, 28, 128, 128, 2),
+                                        (2, 9, 256, 256, 2)]:
+        x = torch.randn(bsz, cin, hgt, hgt, device="cuda").to(
+            memory_format=torch.channels_last).requires_grad_(True)
+        w = (torch.randn(cout, cin, 3, 3, device="cuda") * 0.05
+             ).requires_grad_(True)
+        y = _Conv3x3Fn.apply(x, w, stride)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        y2 = F.conv2d(x2, w2, stride=stride, padding=1)
+        assert y.shape == y2.shape, (y.shape, y2.shape)
+        assert torch.allclose(y, y2, rtol=1e-4, atol=1e-4), \
+            (hgt, cin, cout, stride, (y - y2).abs().max().item())
+        g = torch.randn_like(y).to(memory_format=torch.channels_last)
+        y.backward(g)
+        y2.backward(g)
+        assert torch.allclose(x.grad, x2.grad, rtol=1e-4, atol=1e-4)
+        assert torch.allclose(w.grad, w2.grad, rtol=1e-3, atol=1e-2)
+
+
 def test_smoke_entrypoint():
     import __graft_entry__
     __graft_entry__.smoke()
