@@ -23,6 +23,11 @@ SHAPES = [
     (14, 256, 1024), (14, 1024, 256), (14, 1024, 512),
     (7, 512, 2048), (7, 2048, 512),
 ]
+# (H_in, C, stride) 3x3 shapes in ResNet-50 @224 (Cin == Cout)
+SHAPES_3X3 = [
+    (56, 64, 1), (56, 128, 2), (28, 128, 1), (28, 256, 2),
+    (14, 256, 1), (14, 512, 2), (7, 512, 1),
+]
 
 
 def timeit(fn, iters=10, warmup=3):
@@ -40,7 +45,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--batch", type=int, default=512)
     ap.add_argument("--iters", type=int, default=10)
-    args = ap.parse_args()
+    args, _ = ap.parse_known_args()
     from byol_amd.ops import require_extension
     from byol_amd.ops.conv import _Conv1x1Fn
     require_extension("microbench")
@@ -99,5 +104,23 @@ def main():
               f"{to_b:9.3f}  {ef:.2e} {edx:.2e} {edw:.2e} {tag}")
 
 
+def main_3x3(batch=512, iters=8):
+    from byol_amd.ops.conv import _Conv3x3Fn
+    print(f"{'3x3 shape':>22} {'miopen f':>9} {'mfma f':>9}")
+    for hgt, c, stride in SHAPES_3X3:
+        x = torch.randn(batch, c, hgt, hgt, device="cuda").to(
+            memory_format=torch.channels_last)
+        w = torch.randn(c, c, 3, 3, device="cuda") * 0.05
+        tm = timeit(lambda: F.conv2d(x, w, stride=stride, padding=1), iters)
+        to = timeit(lambda: _Conv3x3Fn.apply(x, w, stride), iters)
+        err = (F.conv2d(x, w, stride=stride, padding=1)
+               - _Conv3x3Fn.apply(x, w, stride)).abs().max().item()
+        tag = "<<" if to < tm else ""
+        print(f"H{hgt:>3} C{c:>5} s{stride} {tm:9.3f} {to:9.3f}  "
+              f"err={err:.2e} {tag}")
+
+
 if __name__ == "__main__":
     main()
+    if "--with-3x3" in sys.argv or os.environ.get("BENCH_3X3"):
+        main_3x3()
