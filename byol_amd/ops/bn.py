@@ -47,12 +47,21 @@ def _rows_view(t: torch.Tensor, c: int) -> torch.Tensor:
 
 
 def _hip_eligible(x: torch.Tensor, c: int) -> bool:
-    if not (x.is_cuda and x.dtype == torch.float32 and c % 4 == 0
-            and has_extension()):
+    if not (x.is_cuda and x.dtype == torch.float32 and c % 4 == 0):
         return False
-    if x.dim() == 4:
-        return x.is_contiguous(memory_format=torch.channels_last)
-    return x.dim() == 2 and x.is_contiguous()
+    if x.dim() == 4 and not x.is_contiguous(
+            memory_format=torch.channels_last):
+        return False
+    if x.dim() == 2 and not x.is_contiguous():
+        return False
+    if x.dim() not in (2, 4):
+        return False
+    # layout/dtype qualify for the HIP path: the extension is now
+    # REQUIRED — a silent ATen fallback on a GPU box would hide a broken
+    # build (fail-loud policy, byol_amd.ops)
+    from . import require_extension
+    require_extension("fused BatchNorm")
+    return True
 
 
 class _FusedBNFunction(torch.autograd.Function):
