@@ -118,6 +118,100 @@ class FolderImages(Dataset):
         return img, label
 
 
+class Cifar10Files(Dataset):
+    """CIFAR-10 from the standard local archives (python-pickle batches
+    ``data_batch_*``/``test_batch`` under ``cifar-10-batches-py``, or the
+    binary ``*.bin`` under ``cifar-10-batches-bin``).  The reference's
+    datasets submodule handled 'simple datasets like MNIST/CIFAR10'
+    (``--download`` flag help, /root/reference/main.py:43-44); there is no
+    network here, so files must already be on disk."""
+
+    def __init__(self, root: str, train: bool):
+        import glob
+        import pickle
+        py_dir = os.path.join(root, "cifar-10-batches-py")
+        bin_dir = os.path.join(root, "cifar-10-batches-bin")
+        imgs, labels = [], []
+        if os.path.isdir(py_dir):
+            names = ([f"data_batch_{i}" for i in range(1, 6)] if train
+                     else ["test_batch"])
+            for nm in names:
+                path = os.path.join(py_dir, nm)
+                if not os.path.isfile(path):
+                    continue
+                with open(path, "rb") as f:
+                    d = pickle.load(f, encoding="bytes")
+                data = d.get(b"data", d.get("data"))
+                labs = d.get(b"labels", d.get("labels"))
+                imgs.append(np.asarray(data, dtype=np.uint8))
+                labels.extend(labs)
+        elif os.path.isdir(bin_dir):
+            names = ([f"data_batch_{i}.bin" for i in range(1, 6)] if train
+                     else ["test_batch.bin"])
+            for nm in names:
+                path = os.path.join(bin_dir, nm)
+                if not os.path.isfile(path):
+                    continue
+                raw = np.fromfile(path, dtype=np.uint8).reshape(-1, 3073)
+                labels.extend(raw[:, 0].tolist())
+                imgs.append(raw[:, 1:])
+        if not imgs:
+            raise FileNotFoundError(
+                f"no CIFAR-10 batches under {py_dir} or {bin_dir}")
+        self.data = np.concatenate(imgs).reshape(-1, 3, 32, 32)
+        self.labels = labels
+        self.num_classes = 10
+
+    def __len__(self):
+        return len(self.labels)
+
+    def __getitem__(self, idx):
+        img = torch.from_numpy(self.data[idx].copy()).float().div_(255.0)
+        return img, int(self.labels[idx])
+
+
+class MnistFiles(Dataset):
+    """MNIST idx-format files (``train-images-idx3-ubyte`` etc., optionally
+    ``.gz``) from ``data_dir`` (or its ``MNIST/raw`` subdir)."""
+
+    def __init__(self, root: str, train: bool):
+        import gzip
+        prefix = "train" if train else "t10k"
+        dirs = [root, os.path.join(root, "MNIST", "raw")]
+        img_path = lab_path = None
+        for d in dirs:
+            for ext in ("", ".gz"):
+                ip = os.path.join(d, f"{prefix}-images-idx3-ubyte{ext}")
+                lp = os.path.join(d, f"{prefix}-labels-idx1-ubyte{ext}")
+                if os.path.isfile(ip) and os.path.isfile(lp):
+                    img_path, lab_path = ip, lp
+        if img_path is None:
+            raise FileNotFoundError(f"no MNIST idx files under {dirs}")
+
+        def read(path):
+            op = gzip.open if path.endswith(".gz") else open
+            with op(path, "rb") as f:
+                return f.read()
+
+        ib = read(img_path)
+        n = int.from_bytes(ib[4:8], "big")
+        h = int.from_bytes(ib[8:12], "big")
+        w = int.from_bytes(ib[12:16], "big")
+        self.data = np.frombuffer(ib, dtype=np.uint8,
+                                  offset=16).reshape(n, h, w)
+        lb = read(lab_path)
+        self.labels = np.frombuffer(lb, dtype=np.uint8, offset=8)
+        self.num_classes = 10
+
+    def __len__(self):
+        return len(self.labels)
+
+    def __getitem__(self, idx):
+        img = torch.from_numpy(self.data[idx].copy()).float().div_(255.0)
+        img = img.unsqueeze(0).repeat(3, 1, 1)  # 3-channel for the encoder
+        return img, int(self.labels[idx])
+
+
 class _SourceDataset(Dataset):
     """Raw (image, label) pairs resized to a fixed source size — what the
     GPU augmentation pipeline consumes (decode/resize on CPU workers, all
@@ -215,6 +309,14 @@ def get_loader(train_transform=None, test_transform=None, **kwargs):
         train_base = FolderImages(os.path.join(data_dir, "train"))
         test_base = FolderImages(os.path.join(data_dir, "test"))
         output_size = train_base.num_classes
+    elif "cifar10" in task:
+        train_base = Cifar10Files(data_dir, train=True)
+        test_base = Cifar10Files(data_dir, train=False)
+        output_size = 10
+    elif "mnist" in task:
+        train_base = MnistFiles(data_dir, train=True)
+        test_base = MnistFiles(data_dir, train=False)
+        output_size = 10
     else:
         raise ValueError(f"unknown task {task!r}")
 

@@ -78,13 +78,20 @@ class Bottleneck(nn.Module):
     expansion = 4
 
     def __init__(self, inplanes: int, planes: int, stride: int = 1,
-                 downsample: Optional[nn.Module] = None):
+                 downsample: Optional[nn.Module] = None, groups: int = 1,
+                 base_width: int = 64):
         super().__init__()
-        self.conv1 = conv1x1(inplanes, planes)
-        self.bn1 = FusedBatchNorm(planes, relu=True)
-        self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = FusedBatchNorm(planes, relu=True)
-        self.conv3 = conv1x1(planes, planes * self.expansion)
+        width = int(planes * (base_width / 64.0)) * groups
+        self.conv1 = conv1x1(inplanes, width)
+        self.bn1 = FusedBatchNorm(width, relu=True)
+        if groups == 1:
+            self.conv2 = conv3x3(width, width, stride)
+        else:  # grouped 3x3 (ResNeXt): plain MIOpen path
+            self.conv2 = nn.Conv2d(width, width, kernel_size=3,
+                                   stride=stride, padding=1, groups=groups,
+                                   bias=False)
+        self.bn2 = FusedBatchNorm(width, relu=True)
+        self.conv3 = conv1x1(width, planes * self.expansion)
         self.bn3 = FusedBatchNorm(planes * self.expansion,
                                   relu=True)  # fused add+relu
         self.downsample = downsample
@@ -97,13 +104,18 @@ class Bottleneck(nn.Module):
         return self.bn3(self.conv3(out), residual=identity)
 
 
+# arch -> (block, layers, groups, base_width); matches torchvision's specs
 ARCH_SPECS = {
-    "resnet18": (BasicBlock, [2, 2, 2, 2]),
-    "resnet34": (BasicBlock, [3, 4, 6, 3]),
-    "resnet50": (Bottleneck, [3, 4, 6, 3]),
-    "resnet101": (Bottleneck, [3, 4, 23, 3]),
-    "resnet152": (Bottleneck, [3, 8, 36, 3]),
-    "resnet200": (Bottleneck, [3, 24, 36, 3]),
+    "resnet18": (BasicBlock, [2, 2, 2, 2], 1, 64),
+    "resnet34": (BasicBlock, [3, 4, 6, 3], 1, 64),
+    "resnet50": (Bottleneck, [3, 4, 6, 3], 1, 64),
+    "resnet101": (Bottleneck, [3, 4, 23, 3], 1, 64),
+    "resnet152": (Bottleneck, [3, 8, 36, 3], 1, 64),
+    "resnet200": (Bottleneck, [3, 24, 36, 3], 1, 64),
+    "wide_resnet50_2": (Bottleneck, [3, 4, 6, 3], 1, 128),
+    "wide_resnet101_2": (Bottleneck, [3, 4, 23, 3], 1, 128),
+    "resnext50_32x4d": (Bottleneck, [3, 4, 6, 3], 32, 4),
+    "resnext101_32x8d": (Bottleneck, [3, 4, 23, 3], 32, 8),
 }
 
 
@@ -120,8 +132,11 @@ class ResNetEncoder(nn.Module):
     """
 
     def __init__(self, block: Type[Union[BasicBlock, Bottleneck]],
-                 layers: List[int], in_channels: int = 3):
+                 layers: List[int], in_channels: int = 3, groups: int = 1,
+                 base_width: int = 64):
         super().__init__()
+        self.groups = groups
+        self.base_width = base_width
         self.inplanes = 64
         self.conv1 = nn.Conv2d(in_channels, 64, kernel_size=7, stride=2,
                                padding=3, bias=False)
@@ -148,9 +163,13 @@ class ResNetEncoder(nn.Module):
         if stride != 1 or self.inplanes != planes * block.expansion:
             downsample = Downsample(self.inplanes,
                                     planes * block.expansion, stride)
-        layers = [block(self.inplanes, planes, stride, downsample)]
+        kw = {}
+        if block is Bottleneck:
+            kw = dict(groups=self.groups, base_width=self.base_width)
+        layers = [block(self.inplanes, planes, stride, downsample, **kw)]
         self.inplanes = planes * block.expansion
-        layers += [block(self.inplanes, planes) for _ in range(1, blocks)]
+        layers += [block(self.inplanes, planes, **kw)
+                   for _ in range(1, blocks)]
         return nn.Sequential(*layers)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -167,5 +186,6 @@ def build_encoder(arch: str, in_channels: int = 3) -> ResNetEncoder:
     if arch not in ARCH_SPECS:
         raise ValueError(
             f"unknown arch {arch!r}; available: {', '.join(arch_names())}")
-    block, layers = ARCH_SPECS[arch]
-    return ResNetEncoder(block, layers, in_channels=in_channels)
+    block, layers, groups, base_width = ARCH_SPECS[arch]
+    return ResNetEncoder(block, layers, in_channels=in_channels,
+                         groups=groups, base_width=base_width)

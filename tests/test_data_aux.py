@@ -129,3 +129,55 @@ def test_main_cli_end_to_end(tmp_path):
                          cwd=ROOT)
     assert res.returncode == 0, res.stderr[-2000:]
     assert "train-0[Epoch 1]" in res.stdout
+
+
+def test_cifar10_task(tmp_path):
+    import pickle
+    from byol_amd.data.loader import get_loader
+    from byol_amd.data.transforms import Resize
+    d = tmp_path / "cifar-10-batches-py"
+    d.mkdir()
+    rng = np.random.RandomState(0)
+    for name, n in [("data_batch_1", 20), ("test_batch", 10)]:
+        payload = {b"data": rng.randint(0, 256, (n, 3072), dtype=np.uint8),
+                   b"labels": rng.randint(0, 10, n).tolist()}
+        with open(d / name, "wb") as f:
+            pickle.dump(payload, f)
+    loader = get_loader(
+        train_transform=[Resize((16, 16))], test_transform=[Resize((16, 16))],
+        task="multi_augment_cifar10", batch_size=4, data_dir=str(tmp_path),
+        num_replicas=1, distributed_rank=0, workers_per_replica=0, seed=0,
+        image_size_override=16, cuda=False)
+    assert loader.output_size == 10
+    assert loader.num_train_samples == 20
+    a1, a2, labels = next(iter(loader.train_loader))
+    assert a1.shape == (4, 3, 16, 16)
+    assert a1.min() >= 0 and a1.max() <= 1
+
+
+def test_mnist_task(tmp_path):
+    from byol_amd.data.loader import get_loader
+    from byol_amd.data.transforms import Resize
+    rng = np.random.RandomState(1)
+
+    def write_idx(prefix, n):
+        imgs = rng.randint(0, 256, (n, 28, 28), dtype=np.uint8)
+        labs = rng.randint(0, 10, n, dtype=np.uint8)
+        with open(tmp_path / f"{prefix}-images-idx3-ubyte", "wb") as f:
+            f.write((2051).to_bytes(4, "big") + n.to_bytes(4, "big")
+                    + (28).to_bytes(4, "big") + (28).to_bytes(4, "big")
+                    + imgs.tobytes())
+        with open(tmp_path / f"{prefix}-labels-idx1-ubyte", "wb") as f:
+            f.write((2049).to_bytes(4, "big") + n.to_bytes(4, "big")
+                    + labs.tobytes())
+
+    write_idx("train", 16)
+    write_idx("t10k", 8)
+    loader = get_loader(
+        train_transform=[Resize((16, 16))], test_transform=[Resize((16, 16))],
+        task="multi_augment_mnist", batch_size=4, data_dir=str(tmp_path),
+        num_replicas=1, distributed_rank=0, workers_per_replica=0, seed=0,
+        image_size_override=16, cuda=False)
+    assert loader.output_size == 10
+    a1, a2, labels = next(iter(loader.train_loader))
+    assert a1.shape == (4, 3, 16, 16)  # mnist expanded to 3 channels

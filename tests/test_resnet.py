@@ -29,7 +29,20 @@ def test_resnet50_backward():
 
 
 @pytest.mark.parametrize("arch,expansion_out", [
-    ("resnet34", 512), ("resnet101", 2048), ("resnet200", 2048)])
+    ("resnet34", 512), ("resnet101", 2048), ("resnet200", 2048),
+    ("wide_resnet50_2", 2048), ("resnext50_32x4d", 2048)])
 def test_other_archs_build(arch, expansion_out):
     enc = build_encoder(arch)
     assert enc.out_channels == expansion_out
+    out = enc(torch.randn(2, 3, 32, 32))
+    assert out.shape == (2, expansion_out, 1, 1)
+
+
+def test_variant_param_counts_match_torchvision():
+    # torchvision reference counts minus the fc layer
+    want = {"wide_resnet50_2": 68_883_240 - 2_049_000,
+            "resnext50_32x4d": 25_028_904 - 2_049_000}
+    for arch, n_want in want.items():
+        enc = build_encoder(arch)
+        n = sum(p.numel() for p in enc.parameters())
+        assert n == n_want, (arch, n, n_want)
