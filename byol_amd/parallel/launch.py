@@ -38,7 +38,10 @@ def init_multiprocessing_and_cuda(rank: int, args):
 
     args.cuda = not args.no_cuda and torch.cuda.is_available()
     if args.cuda:
-        torch.backends.cudnn.benchmark = True  # MIOpen autotune find-mode
+        # reference parity (cudnn.benchmark=True -> MIOpen exhaustive
+        # find); with MIOPEN_FIND_MODE set (main.py defaults it to NORMAL
+        # + the shipped tuned DB) the fast find path is used instead
+        torch.backends.cudnn.benchmark =             os.environ.get("MIOPEN_FIND_MODE", "") == ""
 
     if args.seed is not None:
         import numpy as np
