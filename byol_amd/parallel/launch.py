@@ -41,7 +41,8 @@ def init_multiprocessing_and_cuda(rank: int, args):
         # reference parity (cudnn.benchmark=True -> MIOpen exhaustive
         # find); with MIOPEN_FIND_MODE set (main.py defaults it to NORMAL
         # + the shipped tuned DB) the fast find path is used instead
-        torch.backends.cudnn.benchmark =             os.environ.get("MIOPEN_FIND_MODE", "") == ""
+        torch.backends.cudnn.benchmark = (
+            os.environ.get("MIOPEN_FIND_MODE", "") == "")
 
     if args.seed is not None:
         import numpy as np
