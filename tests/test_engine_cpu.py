@@ -76,3 +76,21 @@ def test_checkpoint_resume(tmp_path):
     args2 = small_args(tmp_path)
     args2.epochs = 3
     run(0, args2)
+
+
+def test_end_to_end_clip(tmp_path):
+    args = small_args(tmp_path, extra=["--clip", "0.1"])
+    run(0, args)
+
+
+def test_end_to_end_lars_sgd_and_lars_adam(tmp_path):
+    for opt in ("lars_sgd", "lars_adam"):
+        args = small_args(tmp_path, extra=["--optimizer", opt])
+        run(0, args)
+
+
+def test_end_to_end_fixed_schedule_and_weight_init(tmp_path):
+    args = small_args(tmp_path, extra=[
+        "--lr-update-schedule", "fixed", "--warmup", "0",
+        "--weight-initialization", "kaiming_normal"])
+    run(0, args)
