@@ -84,8 +84,17 @@ class _Conv1x1Fn(torch.autograd.Function):
             C = extension()
             dx = torch.empty_like(x)
             C.conv1x1_dgrad(_rows(dy, n), wv, _rows(dx, k), m, n, k)
-            dw = torch.zeros(n, k, device=x.device, dtype=x.dtype)
-            C.conv1x1_wgrad(_rows(dy, n), _rows(x, k), dw, m, n, k)
+            if os.environ.get("BYOL_WGRAD", "atomic") == "v2":
+                # round-2 candidate: partial slabs + reduce (no atomics)
+                nchunks = C.wgrad_nchunks(m, n, k)
+                partial = torch.empty(nchunks * n * k, device=x.device,
+                                      dtype=x.dtype)
+                dw = torch.empty(n, k, device=x.device, dtype=x.dtype)
+                C.conv1x1_wgrad_v2(_rows(dy, n), _rows(x, k), partial, dw,
+                                   m, n, k)
+            else:
+                dw = torch.zeros(n, k, device=x.device, dtype=x.dtype)
+                C.conv1x1_wgrad(_rows(dy, n), _rows(x, k), dw, m, n, k)
             return dx, dw.view(n, k, 1, 1)
         # auto mode: MIOpen backward still wins these shapes (see
         # profiles/r01_conv1x1_microbench.md) — route dgrad/wgrad there

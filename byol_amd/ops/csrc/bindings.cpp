@@ -54,6 +54,10 @@ void launch_conv1x1_dgrad(const float* dy, const float* w, float* dx,
                           int64_t m, int n, int k, hipStream_t stream);
 void launch_conv1x1_wgrad(const float* dy, const float* x, float* dw,
                           int64_t m, int n, int k, hipStream_t stream);
+int conv1x1_wgrad_nchunks(int64_t m, int n, int k);
+void launch_conv1x1_wgrad_partial(const float* dy, const float* x,
+                                  float* partial, float* dw, int64_t m,
+                                  int n, int k, hipStream_t stream);
 void launch_conv3x3_fwd(const float* x, const float* wp, float* y, int b,
                         int hi, int wi, int ho, int wo, int k, int n,
                         int stride, hipStream_t stream);
@@ -289,6 +293,23 @@ void conv1x1_wgrad(torch::Tensor dy, torch::Tensor x, torch::Tensor dw,
                        dw.data_ptr<float>(), m, (int)n, (int)k, stream);
 }
 
+int64_t wgrad_nchunks(int64_t m, int64_t n, int64_t k) {
+  return conv1x1_wgrad_nchunks(m, (int)n, (int)k);
+}
+
+void conv1x1_wgrad_v2(torch::Tensor dy, torch::Tensor x,
+                      torch::Tensor partial, torch::Tensor dw, int64_t m,
+                      int64_t n, int64_t k) {
+  CHECK_IN(dy); CHECK_IN(x); CHECK_IN(partial); CHECK_IN(dw);
+  TORCH_CHECK(partial.numel() >= wgrad_nchunks(m, n, k) * n * k,
+              "partial buffer too small");
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_conv1x1_wgrad_partial(dy.data_ptr<float>(), x.data_ptr<float>(),
+                               partial.data_ptr<float>(),
+                               dw.data_ptr<float>(), m, (int)n, (int)k,
+                               stream);
+}
+
 void conv3x3_fwd(torch::Tensor x, torch::Tensor wp, torch::Tensor y,
                  int64_t b, int64_t hi, int64_t wi, int64_t ho, int64_t wo,
                  int64_t k, int64_t n, int64_t stride) {
@@ -306,6 +327,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("conv1x1_fwd", &conv1x1_fwd);
   mod.def("conv1x1_dgrad", &conv1x1_dgrad);
   mod.def("conv1x1_wgrad", &conv1x1_wgrad);
+  mod.def("conv1x1_wgrad_v2", &conv1x1_wgrad_v2);
+  mod.def("wgrad_nchunks", &wgrad_nchunks);
   mod.def("ce_topk_fwd", &ce_topk_fwd);
   mod.def("ce_bwd", &ce_bwd);
   mod.def("aug_sample", &aug_sample);

@@ -397,6 +397,157 @@ void conv1x1_gemm_fast_kernel(const float* __restrict__ A,
   }
 }
 
+// ---------------------------------------------------------------------------
+// wgrad v2 (round-2 candidate): per-chunk PARTIAL slabs with plain stores +
+// a reduce kernel, replacing the atomic epilogue (64 atomic RMWs per lane
+// were the gap vs MIOpen).  partial layout: [chunk][N][K] fp32.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256)
+void conv1x1_wgrad_partial_kernel(const float* __restrict__ dY,
+                                  const float* __restrict__ X,
+                                  float* __restrict__ partial,
+                                  int64_t M, int N, int K, int64_t chunk) {
+  // one shared object, carved (a second __shared__ would also de-pipeline
+  // any future glds version; and the epilogue reuses it as raw scratch)
+  __shared__ __attribute__((aligned(16)))
+      float smem[BK * (BM + 1) + BK * (BN + 1)];
+#define WG_DY(r, c) smem[(r) * (BM + 1) + (c)]
+#define WG_X(r, c) smem[BK * (BM + 1) + (r) * (BN + 1) + (c)]
+
+  const int ntiles_k = (K + BN - 1) / BN;
+  const int ntiles_n = (N + BM - 1) / BM;
+  const int tile_n = (blockIdx.x / ntiles_k) % ntiles_n;
+  const int tile_k = blockIdx.x % ntiles_k;
+  const int64_t mchunk = blockIdx.x / (ntiles_k * ntiles_n);
+  const int n0 = tile_n * BM;
+  const int k0 = tile_k * BN;
+  const int64_t mstart = mchunk * chunk;
+  const int64_t mend = (mstart + chunk < M) ? mstart + chunk : M;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+
+  f32x16 acc[2][2];
+  #pragma unroll
+  for (int t = 0; t < 2; ++t)
+    #pragma unroll
+    for (int u = 0; u < 2; ++u)
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) acc[t][u][e] = 0.f;
+
+  for (int64_t m0 = mstart; m0 < mend; m0 += BK) {
+    {
+      const int r = threadIdx.x >> 3;
+      const int cbase = (threadIdx.x & 7) * 16;
+      #pragma unroll
+      for (int c4 = 0; c4 < 4; ++c4) {
+        const int c = cbase + c4 * 4;
+        float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+        if (m0 + r < mend && n0 + c + 3 < N) {
+          v = *reinterpret_cast<const float4*>(dY + (m0 + r) * N + n0 + c);
+        } else if (m0 + r < mend) {
+          if (n0 + c < N) v.x = dY[(m0 + r) * N + n0 + c];
+          if (n0 + c + 1 < N) v.y = dY[(m0 + r) * N + n0 + c + 1];
+          if (n0 + c + 2 < N) v.z = dY[(m0 + r) * N + n0 + c + 2];
+        }
+        WG_DY(r, c + 0) = v.x;
+        WG_DY(r, c + 1) = v.y;
+        WG_DY(r, c + 2) = v.z;
+        WG_DY(r, c + 3) = v.w;
+      }
+      #pragma unroll
+      for (int c4 = 0; c4 < 4; ++c4) {
+        const int c = cbase + c4 * 4;
+        float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+        if (m0 + r < mend && k0 + c + 3 < K) {
+          v = *reinterpret_cast<const float4*>(X + (m0 + r) * K + k0 + c);
+        } else if (m0 + r < mend) {
+          if (k0 + c < K) v.x = X[(m0 + r) * K + k0 + c];
+          if (k0 + c + 1 < K) v.y = X[(m0 + r) * K + k0 + c + 1];
+          if (k0 + c + 2 < K) v.z = X[(m0 + r) * K + k0 + c + 2];
+        }
+        WG_X(r, c + 0) = v.x;
+        WG_X(r, c + 1) = v.y;
+        WG_X(r, c + 2) = v.z;
+        WG_X(r, c + 3) = v.w;
+      }
+    }
+    __syncthreads();
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 2) {
+      const int mrow = kk + (lane >> 5);
+      #pragma unroll
+      for (int t = 0; t < 2; ++t) {
+        const float a = WG_DY(mrow, wr + t * 32 + (lane & 31));
+        #pragma unroll
+        for (int u = 0; u < 2; ++u) {
+          const float b = WG_X(mrow, wc + u * 32 + (lane & 31));
+          acc[t][u] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc[t][u],
+                                                           0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: LDS-restage two waves at a time (each needs 64x64 f32 =
+  // 4096 floats; smem holds >= 8192), coalesced dwordx4 plain stores.
+  // Every barrier is executed by ALL waves (uniform control flow).
+  float* slab = partial + mchunk * (int64_t)N * K;
+  for (int half = 0; half < 2; ++half) {
+    __syncthreads();
+    if ((wave >> 1) == half) {
+      float* cw = smem + (wave & 1) * 4096;
+      #pragma unroll
+      for (int t = 0; t < 2; ++t)
+        #pragma unroll
+        for (int u = 0; u < 2; ++u)
+          #pragma unroll
+          for (int e = 0; e < 16; ++e) {
+            const int lrow = t * 32 + (e & 3) + 8 * (e >> 2)
+                             + 4 * (lane >> 5);
+            const int lcol = u * 32 + (lane & 31);
+            cw[lrow * 64 + lcol] = acc[t][u][e];
+          }
+    }
+    __syncthreads();
+    if ((wave >> 1) == half) {
+      const float* cw = smem + (wave & 1) * 4096;
+      #pragma unroll
+      for (int p = 0; p < 16; ++p) {
+        const int lrow = p * 4 + (lane >> 4);
+        const int lcol = (lane & 15) * 4;
+        const int gn = n0 + wr + lrow;
+        const int gk = k0 + wc + lcol;
+        if (gn < N && gk + 3 < K) {
+          *reinterpret_cast<float4*>(&slab[(int64_t)gn * K + gk]) =
+              *reinterpret_cast<const float4*>(&cw[lrow * 64 + lcol]);
+        } else if (gn < N) {
+          for (int e = 0; e < 4 && gk + e < K; ++e)
+            slab[(int64_t)gn * K + gk + e] = cw[lrow * 64 + lcol + e];
+        }
+      }
+    }
+  }
+#undef WG_DY
+#undef WG_X
+}
+
+// dW[N*K] = sum over chunks of partial[chunk][N*K]
+__global__ void conv1x1_wgrad_reduce_kernel(const float* __restrict__ partial,
+                                            float* __restrict__ dw,
+                                            int64_t nk, int nchunks) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nk;
+       i += stride) {
+    float v = 0.f;
+    for (int c = 0; c < nchunks; ++c) v += partial[(int64_t)c * nk + i];
+    dw[i] = v;
+  }
+}
+
 static inline int64_t cdiv(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
 void launch_conv1x1_fwd(const float* x, const float* w, const float* wt,
@@ -423,6 +574,29 @@ void launch_conv1x1_dgrad(const float* dy, const float* w, float* dx,
     hipLaunchKernelGGL((conv1x1_gemm_kernel<false>), dim3((uint32_t)grid),
                        dim3(256), 0, stream, dy, w, dx, m, n, k);
   }
+}
+
+int conv1x1_wgrad_nchunks(int64_t m, int n, int k) {
+  const int64_t tiles = cdiv(n, BM) * cdiv(k, BN);
+  int64_t chunk = 32 * BK;
+  while (cdiv(m, chunk) * tiles > 4096) chunk *= 2;
+  return (int)cdiv(m, chunk);
+}
+
+void launch_conv1x1_wgrad_partial(const float* dy, const float* x,
+                                  float* partial, float* dw, int64_t m,
+                                  int n, int k, hipStream_t stream) {
+  const int64_t tiles = cdiv(n, BM) * cdiv(k, BN);
+  int64_t chunk = 32 * BK;
+  while (cdiv(m, chunk) * tiles > 4096) chunk *= 2;
+  const int nchunks = (int)cdiv(m, chunk);
+  const int64_t grid = tiles * nchunks;
+  hipLaunchKernelGGL(conv1x1_wgrad_partial_kernel, dim3((uint32_t)grid),
+                     dim3(256), 0, stream, dy, x, partial, m, n, k, chunk);
+  const int64_t nk = (int64_t)n * k;
+  hipLaunchKernelGGL(conv1x1_wgrad_reduce_kernel,
+                     dim3(grid_1d(nk, 256)), dim3(256), 0, stream, partial,
+                     dw, nk, nchunks);
 }
 
 void launch_conv1x1_wgrad(const float* dy, const float* x, float* dw,

@@ -384,3 +384,28 @@ def test_mfma_conv3x3_vs_miopen(ext, monkeypatch):
 def test_smoke_entrypoint():
     import __graft_entry__
     __graft_entry__.smoke()
+
+
+@pytest.mark.skipif(
+    __import__("os").environ.get("BYOL_TEST_PENDING", "0") != "1",
+    reason="round-2 candidate kernel; enable with BYOL_TEST_PENDING=1")
+def test_wgrad_v2_matches_oracle(ext, monkeypatch):
+    import torch.nn.functional as F
+    from byol_amd.ops.conv import _Conv1x1Fn
+    monkeypatch.setenv("BYOL_MFMA_CONV1X1", "1")
+    monkeypatch.setenv("BYOL_WGRAD", "v2")
+    torch.manual_seed(21)
+    for bsz, hgt, cin, cout in [(8, 56, 64, 256), (4, 7, 2048, 512),
+                                (4, 14, 96, 224)]:
+        x = torch.randn(bsz, cin, hgt, hgt, device="cuda").to(
+            memory_format=torch.channels_last).requires_grad_(True)
+        w = (torch.randn(cout, cin, 1, 1, device="cuda") * 0.05
+             ).requires_grad_(True)
+        y = _Conv1x1Fn.apply(x, w)
+        g = torch.randn_like(y).to(memory_format=torch.channels_last)
+        y.backward(g)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        F.conv2d(x2, w2).backward(g)
+        assert torch.allclose(w.grad, w2.grad, rtol=1e-3, atol=1e-2), \
+            (hgt, cin, cout, (w.grad - w2.grad).abs().max().item())
