@@ -137,8 +137,19 @@ class _Conv3x3Fn(torch.autograd.Function):
         wp = weight.reshape(n, k, 9).permute(2, 1, 0).contiguous()
         y = torch.empty((b, n, ho, wo), device=x.device, dtype=x.dtype,
                         memory_format=torch.channels_last)
-        C.conv3x3_fwd(_rows(x, k), wp, _rows(y, n), b, hi, wi, ho, wo, k, n,
-                      stride)
+        m = b * ho * wo
+        if (m % 128 == 0 and n % 128 == 0 and k % 32 == 0
+                and os.environ.get("BYOL_CONV3X3_FAST", "1") == "1"):
+            # glds-pipelined fast path over a zero-padded copy (guard-free
+            # taps); the pad costs one extra read+write of x
+            xpad = torch.empty(b * (hi + 2) * (wi + 2) * k,
+                               device=x.device, dtype=x.dtype)
+            C.pad_nhwc(_rows(x, k), xpad, b, hi, wi, k)
+            C.conv3x3_fwd_fast(xpad, wp, _rows(y, n), b, hi, wi, ho, wo,
+                               k, n, stride)
+        else:
+            C.conv3x3_fwd(_rows(x, k), wp, _rows(y, n), b, hi, wi, ho, wo,
+                          k, n, stride)
         ctx.save_for_backward(x, weight)
         ctx.stride = stride
         return y

@@ -61,6 +61,11 @@ void launch_conv1x1_wgrad_partial(const float* dy, const float* x,
 void launch_conv3x3_fwd(const float* x, const float* wp, float* y, int b,
                         int hi, int wi, int ho, int wo, int k, int n,
                         int stride, hipStream_t stream);
+void launch_pad_nhwc(const float* x, float* xp, int b, int hi, int wi,
+                     int c, hipStream_t stream);
+void launch_conv3x3_fwd_fast(const float* xp, const float* wp, float* y,
+                             int b, int hi, int wi, int ho, int wo, int k,
+                             int n, int stride, hipStream_t stream);
 void launch_lars_momentum_step(float* p, const float* g, float* m,
                                float* norm_acc, float* alr,
                                const int64_t* seg_off,
@@ -310,6 +315,25 @@ void conv1x1_wgrad_v2(torch::Tensor dy, torch::Tensor x,
                                stream);
 }
 
+void pad_nhwc(torch::Tensor x, torch::Tensor xp, int64_t b, int64_t hi,
+              int64_t wi, int64_t c) {
+  CHECK_IN(x); CHECK_IN(xp);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_pad_nhwc(x.data_ptr<float>(), xp.data_ptr<float>(), (int)b,
+                  (int)hi, (int)wi, (int)c, stream);
+}
+
+void conv3x3_fwd_fast(torch::Tensor xp, torch::Tensor wp, torch::Tensor y,
+                      int64_t b, int64_t hi, int64_t wi, int64_t ho,
+                      int64_t wo, int64_t k, int64_t n, int64_t stride) {
+  CHECK_IN(xp); CHECK_IN(wp); CHECK_IN(y);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_conv3x3_fwd_fast(xp.data_ptr<float>(), wp.data_ptr<float>(),
+                          y.data_ptr<float>(), (int)b, (int)hi, (int)wi,
+                          (int)ho, (int)wo, (int)k, (int)n, (int)stride,
+                          stream);
+}
+
 void conv3x3_fwd(torch::Tensor x, torch::Tensor wp, torch::Tensor y,
                  int64_t b, int64_t hi, int64_t wi, int64_t ho, int64_t wo,
                  int64_t k, int64_t n, int64_t stride) {
@@ -324,6 +348,8 @@ void conv3x3_fwd(torch::Tensor x, torch::Tensor wp, torch::Tensor y,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("conv3x3_fwd", &conv3x3_fwd);
+  mod.def("conv3x3_fwd_fast", &conv3x3_fwd_fast);
+  mod.def("pad_nhwc", &pad_nhwc);
   mod.def("conv1x1_fwd", &conv1x1_fwd);
   mod.def("conv1x1_dgrad", &conv1x1_dgrad);
   mod.def("conv1x1_wgrad", &conv1x1_wgrad);

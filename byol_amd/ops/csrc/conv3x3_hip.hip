@@ -142,6 +142,195 @@ void conv3x3_fwd_kernel(const float* __restrict__ X,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fast path (round-2 candidate): zero-padded input copy makes every tap an
+// unguarded uniform shift, so A and B both stage through glds
+// (global_load_lds) with a 2-deep double buffer across the (tap, k0)
+// sequence — the conv1x1 fast-path structure with 9x the K depth.
+// Requires M%128==0, N%128==0, K%32==0.
+// ---------------------------------------------------------------------------
+
+__global__ void pad_nhwc_kernel(const float* __restrict__ x,
+                                float* __restrict__ xp,
+                                int b, int hi, int wi, int c) {
+  const int hp = hi + 2, wp_ = wi + 2;
+  const int c4 = c >> 2;
+  const int64_t total = (int64_t)b * hp * wp_ * c4;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const float4* x4 = reinterpret_cast<const float4*>(x);
+  float4* o4 = reinterpret_cast<float4*>(xp);
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    const int q = (int)(i % c4);
+    const int64_t pix = i / c4;
+    const int xw = (int)(pix % wp_);
+    const int yh = (int)((pix / wp_) % hp);
+    const int bb = (int)(pix / ((int64_t)wp_ * hp));
+    float4 v = make_float4(0.f, 0.f, 0.f, 0.f);
+    if (yh >= 1 && yh <= hi && xw >= 1 && xw <= wi) {
+      v = x4[(((int64_t)bb * hi + (yh - 1)) * wi + (xw - 1)) * c4 + q];
+    }
+    o4[i] = v;
+  }
+}
+
+#define C3F_LDSW_A 4096
+#define C3F_LDSW_B 4096
+#define C3F_LDSW_BUF (C3F_LDSW_A + C3F_LDSW_B)
+
+__global__ __launch_bounds__(256)
+void conv3x3_fwd_fast_kernel(const float* __restrict__ Xp,  // padded NHWC
+                             const float* __restrict__ Wp,  // [9][K][N]
+                             float* __restrict__ Y,
+                             int B, int Hp, int Wpp,  // padded dims
+                             int Ho, int Wo, int K, int N, int stride) {
+  __shared__ __attribute__((aligned(16))) float lds[2 * C3F_LDSW_BUF];
+
+  const int64_t M = (int64_t)B * Ho * Wo;
+  const int ntiles_n = N / C3_BN;
+  const int tile_m = blockIdx.x / ntiles_n;
+  const int tile_n = blockIdx.x % ntiles_n;
+  const int64_t m0 = (int64_t)tile_m * C3_BM;
+  const int n0 = tile_n * C3_BN;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+
+  // per-thread A staging geometry: 4 glds issues, each covering LDS words
+  // (wave*4+i)*256 + lane*4 -> row r (of 128), quad q (of 8).  The output
+  // row r is fixed per issue, so its padded-source base is precomputed.
+  const float* abase[4];
+  int aq[4];
+  #pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int off = (wave * 4 + i) * 256 + lane * 4;
+    const int r = off >> 5;
+    const int q = (off & 31) >> 2;
+    aq[i] = ((q - r) & 7) << 2;  // quad-rotated source column (rule 21)
+    const int64_t gm = m0 + r;
+    const int64_t howo = (int64_t)Ho * Wo;
+    const int bb = (int)(gm / howo);
+    const int rem = (int)(gm % howo);
+    const int iy = (rem / Wo) * stride;   // padded coords: no -1, no guard
+    const int ix = (rem % Wo) * stride;
+    abase[i] = Xp + (((int64_t)bb * Hp + iy) * Wpp + ix) * K;
+  }
+
+  f32x16_c3 acc[2][2];
+  #pragma unroll
+  for (int t = 0; t < 2; ++t)
+    #pragma unroll
+    for (int u = 0; u < 2; ++u)
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) acc[t][u][e] = 0.f;
+
+  const int nk = K / C3_BK;
+  const int nsteps = 9 * nk;
+
+  auto stage = [&](int buf, int step) {
+    const int tap = step / nk;
+    const int k0 = (step % nk) * C3_BK;
+    const int dy = tap / 3, dx = tap % 3;
+    const int64_t tap_off = ((int64_t)dy * Wpp + dx) * K;
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const float* src = abase[i] + tap_off + k0 + aq[i];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * C3F_LDSW_BUF + (wave * 4 + i) * 256],
+          16, 0, 0);
+    }
+    const float* wtap = Wp + ((int64_t)tap * K + k0) * N;
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int off = (wave * 4 + i) * 256 + lane * 4;
+      const int r = off >> 7;
+      const int c = off & 127;
+      const float* src = wtap + (int64_t)r * N + n0 + c;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * C3F_LDSW_BUF + C3F_LDSW_A + (wave * 4 + i) * 256],
+          16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int buf = 0;
+  for (int step = 0; step < nsteps; ++step) {
+    if (step + 1 < nsteps) stage(buf ^ 1, step + 1);
+    const float* la = &lds[buf * C3F_LDSW_BUF];
+    const float* lb = &lds[buf * C3F_LDSW_BUF + C3F_LDSW_A];
+    #pragma unroll
+    for (int kk = 0; kk < C3_BK; kk += 2) {
+      const int krow = kk + (lane >> 5);
+      #pragma unroll
+      for (int t = 0; t < 2; ++t) {
+        const int row = wr + t * 32 + (lane & 31);
+        const float a = la[row * 32 + ((((krow >> 2) + row) & 7) << 2)
+                           + (krow & 3)];
+        #pragma unroll
+        for (int u = 0; u < 2; ++u) {
+          const float b = lb[krow * C3_BN + wc + u * 32 + (lane & 31)];
+          acc[t][u] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc[t][u],
+                                                           0, 0, 0);
+        }
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+
+  // LDS-restaged coalesced epilogue (same as conv1x1 fast)
+  __syncthreads();
+  float* cw = &lds[wave * 4096];
+  #pragma unroll
+  for (int t = 0; t < 2; ++t)
+    #pragma unroll
+    for (int u = 0; u < 2; ++u)
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        const int lrow = t * 32 + (e & 3) + 8 * (e >> 2) + 4 * (lane >> 5);
+        const int lcol = u * 32 + (lane & 31);
+        cw[lrow * 64 + lcol] = acc[t][u][e];
+      }
+  __builtin_amdgcn_s_barrier();
+  const int64_t gm_base = m0 + wr;
+  const int gn_base = n0 + wc;
+  #pragma unroll
+  for (int p = 0; p < 16; ++p) {
+    const int lrow = p * 4 + (lane >> 4);
+    const int lcol = (lane & 15) * 4;
+    const float4 v = *reinterpret_cast<const float4*>(&cw[lrow * 64 + lcol]);
+    *reinterpret_cast<float4*>(
+        &Y[(gm_base + lrow) * N + gn_base + lcol]) = v;
+  }
+}
+
+void launch_pad_nhwc(const float* x, float* xp, int b, int hi, int wi,
+                     int c, hipStream_t stream) {
+  const int64_t total = (int64_t)b * (hi + 2) * (wi + 2) * (c >> 2);
+  hipLaunchKernelGGL(pad_nhwc_kernel, dim3(grid_1d(total, 256)), dim3(256),
+                     0, stream, x, xp, b, hi, wi, c);
+}
+
+void launch_conv3x3_fwd_fast(const float* xp, const float* wp, float* y,
+                             int b, int hi, int wi, int ho, int wo, int k,
+                             int n, int stride, hipStream_t stream) {
+  const int64_t m = (int64_t)b * ho * wo;
+  const int64_t grid = (m / C3_BM) * (n / C3_BN);
+  hipLaunchKernelGGL(conv3x3_fwd_fast_kernel, dim3((uint32_t)grid),
+                     dim3(256), 0, stream, xp, wp, y, b, hi + 2, wi + 2, ho,
+                     wo, k, n, stride);
+}
+
 void launch_conv3x3_fwd(const float* x, const float* wp, float* y, int b,
                         int hi, int wi, int ho, int wo, int k, int n,
                         int stride, hipStream_t stream) {

@@ -409,3 +409,27 @@ def test_wgrad_v2_matches_oracle(ext, monkeypatch):
         F.conv2d(x2, w2).backward(g)
         assert torch.allclose(w.grad, w2.grad, rtol=1e-3, atol=1e-2), \
             (hgt, cin, cout, (w.grad - w2.grad).abs().max().item())
+
+
+@pytest.mark.skipif(
+    __import__("os").environ.get("BYOL_TEST_PENDING", "0") != "1",
+    reason="round-2 candidate kernel; enable with BYOL_TEST_PENDING=1")
+def test_conv3x3_fast_path_matches_oracle(ext, monkeypatch):
+    """glds fast path (padded input) — fast-eligible shapes (M%128==0)."""
+    import torch.nn.functional as F
+    from byol_amd.ops.conv import _Conv3x3Fn
+    monkeypatch.setenv("BYOL_CONV3X3_FAST", "1")
+    torch.manual_seed(22)
+    for bsz, hgt, c, stride in [(8, 16, 128, 1), (8, 16, 256, 2),
+                                (2, 32, 512, 1)]:
+        x = torch.randn(bsz, c, hgt, hgt, device="cuda").to(
+            memory_format=torch.channels_last)
+        w = torch.randn(c, c, 3, 3, device="cuda") * 0.05
+        ho = (hgt + 2 - 3) // stride + 1
+        m = bsz * ho * ho
+        if m % 128 != 0:
+            continue
+        y = _Conv3x3Fn.apply(x, w, stride)
+        y2 = F.conv2d(x, w, stride=stride, padding=1)
+        assert torch.allclose(y, y2, rtol=1e-4, atol=1e-4), \
+            (hgt, c, stride, (y - y2).abs().max().item())
