@@ -18,6 +18,7 @@ tests.
 """
 
 import math
+import os
 from typing import Optional, Tuple
 
 import numpy as np
@@ -149,8 +150,9 @@ class GPUTwoViewAugment:
         cparam = torch.from_numpy(cparam_np).to(device, non_blocking=True)
         dst = torch.empty(b, s, s, 3, device=device, dtype=torch.float32)
         gray_sum = torch.zeros(b, device=device, dtype=torch.float32)
+        use_v2 = 1 if os.environ.get("BYOL_AUG_V2", "0") == "1" else 0
         ext.aug_sample(src_nhwc.reshape(-1), dst.reshape(-1), gray_sum,
-                       crop.reshape(-1), hs, ws, s)
+                       crop.reshape(-1), hs, ws, s, use_v2)
         ext.aug_color(dst.reshape(-1), gray_sum, cparam.reshape(-1), s)
         # NHWC [B,S,S,3] -> channels_last NCHW view
         out = dst.permute(0, 3, 1, 2)

@@ -156,9 +156,75 @@ __global__ void aug_color_kernel(float* __restrict__ img,
   }
 }
 
+// v2 (round-2 candidate): sample-major blocks — each block covers a pixel
+// chunk of ONE sample, reduces its gray contribution in LDS and issues a
+// single atomicAdd, instead of one atomic per pixel (s*s serialized RMWs
+// per sample address in v1).
+__global__ void aug_sample_v2_kernel(const float* __restrict__ src,
+                                     float* __restrict__ dst,
+                                     float* __restrict__ gray_sum,
+                                     const float* __restrict__ crop,
+                                     int hs, int ws, int s,
+                                     int chunks_per_sample) {
+  __shared__ float scratch[4];
+  const int bi = blockIdx.x / chunks_per_sample;
+  const int chunk = blockIdx.x % chunks_per_sample;
+  const int npix = s * s;
+  const int per_chunk = (npix + chunks_per_sample - 1) / chunks_per_sample;
+  const int start = chunk * per_chunk;
+  const int end = min(start + per_chunk, npix);
+
+  const float* cp = crop + bi * 5;
+  const float y0 = cp[0], x0 = cp[1], ch = cp[2], cw = cp[3];
+  const int flip = (int)cp[4];
+  const float* sb = src + (int64_t)bi * hs * ws * 3;
+  float* db = dst + (int64_t)bi * npix * 3;
+
+  float gacc = 0.f;
+  for (int p = start + threadIdx.x; p < end; p += blockDim.x) {
+    const int ox = p % s;
+    const int oy = p / s;
+    const int sx = flip ? (s - 1 - ox) : ox;
+    float fy = y0 + ((float)oy + 0.5f) * ch / (float)s - 0.5f;
+    float fx = x0 + ((float)sx + 0.5f) * cw / (float)s - 0.5f;
+    const int y1i = (int)floorf(fy);
+    const int x1i = (int)floorf(fx);
+    const float wy = fy - (float)y1i;
+    const float wx = fx - (float)x1i;
+    const int ya = min(max(y1i, 0), hs - 1);
+    const int yb = min(max(y1i + 1, 0), hs - 1);
+    const int xa = min(max(x1i, 0), ws - 1);
+    const int xb = min(max(x1i + 1, 0), ws - 1);
+    float out[3];
+    #pragma unroll
+    for (int col = 0; col < 3; ++col) {
+      const float v00 = sb[((int64_t)ya * ws + xa) * 3 + col];
+      const float v01 = sb[((int64_t)ya * ws + xb) * 3 + col];
+      const float v10 = sb[((int64_t)yb * ws + xa) * 3 + col];
+      const float v11 = sb[((int64_t)yb * ws + xb) * 3 + col];
+      const float v0 = v00 + wx * (v01 - v00);
+      const float v1 = v10 + wx * (v11 - v10);
+      float v = v0 + wy * (v1 - v0);
+      v = fminf(fmaxf(v, 0.f), 1.f);
+      out[col] = v;
+      db[(int64_t)p * 3 + col] = v;
+    }
+    gacc += GRAY_R * out[0] + GRAY_G * out[1] + GRAY_B * out[2];
+  }
+  const float tot = block_reduce_sum(gacc, scratch);
+  if (threadIdx.x == 0) atomicAdd(&gray_sum[bi], tot);
+}
+
 void launch_aug_sample(const float* src, float* dst, float* gray_sum,
                        const float* crop, int b, int hs, int ws, int s,
-                       hipStream_t stream) {
+                       int use_v2, hipStream_t stream) {
+  if (use_v2) {  // round-2 candidate: sample-major, one atomic per block
+    int chunks = (s * s + 8191) / 8192;
+    if (chunks < 1) chunks = 1;
+    hipLaunchKernelGGL(aug_sample_v2_kernel, dim3(b * chunks), dim3(256), 0,
+                       stream, src, dst, gray_sum, crop, hs, ws, s, chunks);
+    return;
+  }
   const int64_t total = (int64_t)b * s * s;
   hipLaunchKernelGGL(aug_sample_kernel, dim3(grid_1d(total, 256)), dim3(256),
                      0, stream, src, dst, gray_sum, crop, b, hs, ws, s);

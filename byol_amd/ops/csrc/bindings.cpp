@@ -38,7 +38,7 @@ void launch_bn_bwd_apply(const float* dy, const float* y, const float* x,
                          int relu, hipStream_t stream);
 void launch_aug_sample(const float* src, float* dst, float* gray_sum,
                        const float* crop, int b, int hs, int ws, int s,
-                       hipStream_t stream);
+                       int use_v2, hipStream_t stream);
 void launch_aug_color(float* img, const float* gray_sum, const float* cparam,
                       int b, int s, hipStream_t stream);
 void launch_ce_topk_fwd(const float* logits, const int64_t* labels,
@@ -233,13 +233,14 @@ void bn_bwd_apply(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
 }
 
 void aug_sample(torch::Tensor src, torch::Tensor dst, torch::Tensor gray_sum,
-                torch::Tensor crop, int64_t hs, int64_t ws, int64_t s) {
+                torch::Tensor crop, int64_t hs, int64_t ws, int64_t s,
+                int64_t use_v2) {
   CHECK_IN(src); CHECK_IN(dst); CHECK_IN(gray_sum); CHECK_IN(crop);
   const int b = gray_sum.numel();
   auto stream = at::hip::getCurrentHIPStream();
   launch_aug_sample(src.data_ptr<float>(), dst.data_ptr<float>(),
                     gray_sum.data_ptr<float>(), crop.data_ptr<float>(), b,
-                    (int)hs, (int)ws, (int)s, stream);
+                    (int)hs, (int)ws, (int)s, (int)use_v2, stream);
 }
 
 void aug_color(torch::Tensor img, torch::Tensor gray_sum,

@@ -279,7 +279,7 @@ def test_aug_sample_matches_interpolate(ext):
     dst = torch.empty(b, s, s, 3, device="cuda")
     gray = torch.zeros(b, device="cuda")
     ext.aug_sample(src_nhwc.reshape(-1), dst.reshape(-1), gray, crop,
-                   hs, ws, s)
+                   hs, ws, s, 0)
     want = F.interpolate(src, size=(s, s), mode="bilinear",
                          align_corners=False)
     got = dst.permute(0, 3, 1, 2)
@@ -433,3 +433,25 @@ def test_conv3x3_fast_path_matches_oracle(ext, monkeypatch):
         y2 = F.conv2d(x, w, stride=stride, padding=1)
         assert torch.allclose(y, y2, rtol=1e-4, atol=1e-4), \
             (hgt, c, stride, (y - y2).abs().max().item())
+
+
+@pytest.mark.skipif(
+    __import__("os").environ.get("BYOL_TEST_PENDING", "0") != "1",
+    reason="round-2 candidate kernel; enable with BYOL_TEST_PENDING=1")
+def test_aug_sample_v2_matches_v1(ext):
+    import torch.nn.functional as F
+    torch.manual_seed(14)
+    b, hs, ws, s = 6, 70, 50, 32
+    src = torch.rand(b, hs, ws, 3, device="cuda").contiguous()
+    crop = torch.tensor(
+        [[3.0, 2.0, 40, 30, float(i % 2)] for i in range(b)],
+        device="cuda").reshape(-1)
+    out = []
+    for v2 in (0, 1):
+        dst = torch.empty(b, s, s, 3, device="cuda")
+        gray = torch.zeros(b, device="cuda")
+        ext.aug_sample(src.reshape(-1), dst.reshape(-1), gray, crop,
+                       hs, ws, s, v2)
+        out.append((dst, gray))
+    assert torch.equal(out[0][0], out[1][0])
+    assert torch.allclose(out[0][1], out[1][1], rtol=1e-4, atol=1e-3)
