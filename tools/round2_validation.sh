@@ -10,7 +10,7 @@ timeout 400 python -m pytest tests -q -m gpu 2>&1 | tail -3
 
 # 2. pending candidates: wgrad v2 + conv3x3 glds fast path
 BYOL_TEST_PENDING=1 timeout 300 python -m pytest tests/test_ops_gpu.py -q \
-    -m gpu -k "wgrad_v2 or conv3x3_fast" 2>&1 | tail -3
+    -m gpu -k "wgrad_v2 or conv3x3_fast or aug_sample_v2" 2>&1 | tail -3
 
 # 3. per-shape measurements (1x1 incl. BYOL_WGRAD=v2 A/B, 3x3 fast/slow A/B)
 timeout 400 python tools/conv_microbench.py --batch 512 --iters 8 2>&1 | tail -16
