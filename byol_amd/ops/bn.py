@@ -21,6 +21,7 @@ i.e. 4-D channels_last tensors or plain 2-D [B, C] (the projector/predictor
 BatchNorm1d — same kernels).
 """
 
+import os
 from typing import Optional
 
 import torch
@@ -46,8 +47,15 @@ def _rows_view(t: torch.Tensor, c: int) -> torch.Tensor:
     return t.reshape(-1, c)
 
 
+def _bf16_enabled() -> bool:
+    # round-2 candidate path: bf16-I/O BN kernels (BYOL_BF16_BN=1)
+    return os.environ.get("BYOL_BF16_BN", "0") == "1"
+
+
 def _hip_eligible(x: torch.Tensor, c: int) -> bool:
-    if not (x.is_cuda and x.dtype == torch.float32 and c % 4 == 0):
+    ok_dtype = (x.dtype == torch.float32
+                or (x.dtype == torch.bfloat16 and _bf16_enabled()))
+    if not (x.is_cuda and ok_dtype and c % 4 == 0):
         return False
     if x.dim() == 4 and not x.is_contiguous(
             memory_format=torch.channels_last):
@@ -74,11 +82,15 @@ class _FusedBNFunction(torch.autograd.Function):
         m = xv.shape[0]
         world = _world(process_group) if sync else 1
 
+        bf16 = x.dtype == torch.bfloat16
         if training:
             nslots = 64  # atomic-contention fanout (see csrc/bn_fused.hip)
             acc = torch.zeros(nslots * 2 * c, device=x.device,
                               dtype=torch.float32)
-            C.bn_stats(xv, acc, m, c, nslots)
+            if bf16:
+                C.bn_stats_bf16(xv, acc, m, c, nslots)
+            else:
+                C.bn_stats(xv, acc, m, c, nslots)
             acc2c = torch.empty(2 * c, device=x.device, dtype=torch.float32)
             C.bn_reduce_slots(acc, acc2c, nslots)
             count = float(m)
@@ -98,8 +110,12 @@ class _FusedBNFunction(torch.autograd.Function):
         y = torch.empty_like(x)
         yv = _rows_view(y, c)
         rv = _rows_view(residual, c) if residual is not None else None
-        C.bn_apply(xv, rv, mean, invstd, weight, bias, yv, m, c,
-                   1 if relu else 0)
+        if bf16:
+            C.bn_apply_bf16(xv, rv, mean, invstd, weight, bias, yv, m, c,
+                            1 if relu else 0)
+        else:
+            C.bn_apply(xv, rv, mean, invstd, weight, bias, yv, m, c,
+                       1 if relu else 0)
         ctx.save_for_backward(x, y, weight, mean, invstd)
         ctx.relu = relu
         ctx.has_residual = residual is not None
@@ -114,6 +130,9 @@ class _FusedBNFunction(torch.autograd.Function):
         C = extension()
         x, y, weight, mean, invstd = ctx.saved_tensors
         c = ctx.c
+        bf16 = x.dtype == torch.bfloat16
+        if bf16 and dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
         dy = dy.contiguous(memory_format=torch.channels_last) \
             if dy.dim() == 4 else dy.contiguous()
         dyv = _rows_view(dy, c)
@@ -123,8 +142,12 @@ class _FusedBNFunction(torch.autograd.Function):
         nslots = 64
         red_s = torch.zeros(nslots * 2 * c, device=x.device,
                             dtype=torch.float32)
-        C.bn_bwd_reduce(dyv, yv, xv, mean, invstd, red_s, m, c,
-                        1 if ctx.relu else 0, nslots)
+        if bf16:
+            C.bn_bwd_reduce_bf16(dyv, yv, xv, mean, invstd, red_s, m, c,
+                                 1 if ctx.relu else 0, nslots)
+        else:
+            C.bn_bwd_reduce(dyv, yv, xv, mean, invstd, red_s, m, c,
+                            1 if ctx.relu else 0, nslots)
         red = torch.empty(2 * c, device=x.device, dtype=torch.float32)
         C.bn_reduce_slots(red_s, red, nslots)
         # weight/bias grads are LOCAL sums (DDP averages them afterwards,
@@ -140,8 +163,13 @@ class _FusedBNFunction(torch.autograd.Function):
         if ctx.has_residual:
             dres = torch.empty_like(x)
             drv = _rows_view(dres, c)
-        C.bn_bwd_apply(dyv, yv, xv, mean, invstd, weight, red, dxv, drv,
-                       1.0 / ctx.count, m, c, 1 if ctx.relu else 0)
+        if bf16:
+            C.bn_bwd_apply_bf16(dyv, yv, xv, mean, invstd, weight, red,
+                                dxv, drv, 1.0 / ctx.count, m, c,
+                                1 if ctx.relu else 0)
+        else:
+            C.bn_bwd_apply(dyv, yv, xv, mean, invstd, weight, red, dxv,
+                           drv, 1.0 / ctx.count, m, c, 1 if ctx.relu else 0)
         return (dx, dres, dw, db, None, None, None, None, None, None, None,
                 None)
 

@@ -31,6 +31,25 @@ void launch_bn_bwd_reduce(const float* dy, const float* y, const float* x,
                           const float* mean, const float* invstd, float* red,
                           int64_t m, int c, int relu, int slot_mask,
                           hipStream_t stream);
+void launch_bn_stats_bf16(const uint16_t* x, float* acc, int64_t m, int c,
+                          int slot_mask, hipStream_t stream);
+void launch_bn_apply_bf16(const uint16_t* x, const uint16_t* residual,
+                          const float* mean, const float* invstd,
+                          const float* weight, const float* bias,
+                          uint16_t* y, int64_t m, int c, int relu,
+                          hipStream_t stream);
+void launch_bn_bwd_reduce_bf16(const uint16_t* dy, const uint16_t* y,
+                               const uint16_t* x, const float* mean,
+                               const float* invstd, float* red, int64_t m,
+                               int c, int relu, int slot_mask,
+                               hipStream_t stream);
+void launch_bn_bwd_apply_bf16(const uint16_t* dy, const uint16_t* y,
+                              const uint16_t* x, const float* mean,
+                              const float* invstd, const float* weight,
+                              const float* red, uint16_t* dx,
+                              uint16_t* dresidual, float inv_count,
+                              int64_t m, int c, int relu,
+                              hipStream_t stream);
 void launch_bn_bwd_apply(const float* dy, const float* y, const float* x,
                          const float* mean, const float* invstd,
                          const float* weight, const float* red, float* dx,
@@ -345,9 +364,79 @@ void conv3x3_fwd(torch::Tensor x, torch::Tensor wp, torch::Tensor y,
                      (int)wo, (int)k, (int)n, (int)stride, stream);
 }
 
+
+// --- bf16-I/O fused BN (round-2 candidate; stats/params stay fp32) --------
+
+#define CHECK_BF(t)                                                        \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                        \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous");              \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16")
+
+static const uint16_t* bfp(const torch::Tensor& t) {
+  return reinterpret_cast<const uint16_t*>(t.data_ptr<at::BFloat16>());
+}
+static uint16_t* bfp_mut(torch::Tensor& t) {
+  return reinterpret_cast<uint16_t*>(t.data_ptr<at::BFloat16>());
+}
+
+void bn_stats_bf16(torch::Tensor x, torch::Tensor acc, int64_t m, int64_t c,
+                   int64_t nslots) {
+  CHECK_BF(x); CHECK_IN(acc);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_stats_bf16(bfp(x), acc.data_ptr<float>(), m, (int)c,
+                       (int)(nslots - 1), stream);
+}
+
+void bn_apply_bf16(torch::Tensor x, c10::optional<torch::Tensor> residual,
+                   torch::Tensor mean, torch::Tensor invstd,
+                   torch::Tensor weight, torch::Tensor bias, torch::Tensor y,
+                   int64_t m, int64_t c, int64_t relu) {
+  CHECK_BF(x); CHECK_BF(y);
+  CHECK_IN(mean); CHECK_IN(invstd); CHECK_IN(weight); CHECK_IN(bias);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_apply_bf16(bfp(x), residual ? bfp(*residual) : nullptr,
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       weight.data_ptr<float>(), bias.data_ptr<float>(),
+                       bfp_mut(y), m, (int)c, (int)relu, stream);
+}
+
+void bn_bwd_reduce_bf16(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+                        torch::Tensor mean, torch::Tensor invstd,
+                        torch::Tensor red, int64_t m, int64_t c,
+                        int64_t relu, int64_t nslots) {
+  CHECK_BF(dy); CHECK_BF(y); CHECK_BF(x); CHECK_IN(red);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_bwd_reduce_bf16(bfp(dy), bfp(y), bfp(x),
+                            mean.data_ptr<float>(),
+                            invstd.data_ptr<float>(), red.data_ptr<float>(),
+                            m, (int)c, (int)relu, (int)(nslots - 1), stream);
+}
+
+void bn_bwd_apply_bf16(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+                       torch::Tensor mean, torch::Tensor invstd,
+                       torch::Tensor weight, torch::Tensor red,
+                       torch::Tensor dx, c10::optional<torch::Tensor> dres,
+                       double inv_count, int64_t m, int64_t c,
+                       int64_t relu) {
+  CHECK_BF(dy); CHECK_BF(y); CHECK_BF(x); CHECK_BF(dx);
+  auto stream = at::hip::getCurrentHIPStream();
+  torch::Tensor drt;
+  uint16_t* drp = nullptr;
+  if (dres) { drt = *dres; drp = bfp_mut(drt); }
+  launch_bn_bwd_apply_bf16(bfp(dy), bfp(y), bfp(x), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(),
+                           weight.data_ptr<float>(), red.data_ptr<float>(),
+                           bfp_mut(dx), drp, (float)inv_count, m, (int)c,
+                           (int)relu, stream);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("bn_stats_bf16", &bn_stats_bf16);
+  mod.def("bn_apply_bf16", &bn_apply_bf16);
+  mod.def("bn_bwd_reduce_bf16", &bn_bwd_reduce_bf16);
+  mod.def("bn_bwd_apply_bf16", &bn_bwd_apply_bf16);
   mod.def("conv3x3_fwd", &conv3x3_fwd);
   mod.def("conv3x3_fwd_fast", &conv3x3_fwd_fast);
   mod.def("pad_nhwc", &pad_nhwc);

@@ -24,6 +24,7 @@ sources = [
     os.path.join(CSRC, "cetopk.hip"),
     os.path.join(CSRC, "conv1x1.hip"),
     os.path.join(CSRC, "conv3x3.hip"),
+    os.path.join(CSRC, "bn_bf16.hip"),
 ]
 
 setup(

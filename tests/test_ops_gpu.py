@@ -455,3 +455,45 @@ def test_aug_sample_v2_matches_v1(ext):
         out.append((dst, gray))
     assert torch.equal(out[0][0], out[1][0])
     assert torch.allclose(out[0][1], out[1][1], rtol=1e-4, atol=1e-3)
+
+
+@pytest.mark.skipif(
+    __import__("os").environ.get("BYOL_TEST_PENDING", "0") != "1",
+    reason="round-2 candidate kernel; enable with BYOL_TEST_PENDING=1")
+def test_bf16_fused_bn_vs_fp32_oracle(ext, monkeypatch):
+    """bf16-I/O BN kernels vs an fp32 reference on the same (bf16-rounded)
+    inputs; stats/params stay fp32 so tolerances are bf16-IO-class."""
+    import torch.nn as nn
+    from byol_amd.ops.bn import FusedBatchNorm
+    monkeypatch.setenv("BYOL_BF16_BN", "1")
+    torch.manual_seed(15)
+    for shape, relu, residual in [((16, 64, 14, 14), True, False),
+                                  ((8, 256, 7, 7), True, True)]:
+        c = shape[1]
+        fused = FusedBatchNorm(c, relu=relu).cuda()
+        ref = nn.BatchNorm2d(c).cuda()
+        xb = torch.randn(shape, device="cuda").to(torch.bfloat16)
+        x1 = xb.to(memory_format=torch.channels_last).requires_grad_(True)
+        x2 = xb.float().clone().requires_grad_(True)
+        r1 = r2 = None
+        if residual:
+            rb = torch.randn(shape, device="cuda").to(torch.bfloat16)
+            r1 = rb.to(memory_format=torch.channels_last).requires_grad_(True)
+            r2 = rb.float().clone().requires_grad_(True)
+        y1 = fused(x1, residual=r1)
+        assert y1.dtype == torch.bfloat16
+        y2 = ref(x2)
+        if residual:
+            y2 = y2 + r2
+        if relu:
+            y2 = torch.relu(y2)
+        assert torch.allclose(y1.float(), y2, rtol=2e-2, atol=2e-2), \
+            (y1.float() - y2).abs().max().item()
+        g = torch.randn_like(y2).to(torch.bfloat16)
+        y1.backward(g)
+        y2.backward(g.float())
+        assert torch.allclose(x1.grad.float(), x2.grad, rtol=5e-2,
+                              atol=5e-2), \
+            (x1.grad.float() - x2.grad).abs().max().item()
+        assert torch.allclose(fused.running_mean, ref.running_mean,
+                              atol=5e-3)
