@@ -79,7 +79,8 @@ class CheckpointBundle:
             self.optimizer.load_state_dict(payload["optimizer"])
         if self.scheduler is not None and payload.get("scheduler"):
             self.scheduler.load_state_dict(payload["scheduler"])
-        return {"epoch": payload.get("epoch", 0) + 1}
+        return {"epoch": payload.get("epoch", 0) + 1,
+                "best_loss": payload.get("best_loss")}
 
 
 class ModelSaver:
@@ -100,6 +101,10 @@ class ModelSaver:
     def restore(self) -> dict:
         state = self.bundle.load()
         self.epoch = state["epoch"] - 1
+        # re-seed best from the checkpoint so a resumed run does not
+        # overwrite the best checkpoint with a worse post-restart epoch
+        if state.get("best_loss") is not None:
+            self.best = state["best_loss"]
         return state
 
     def __call__(self, loss: float) -> bool:

@@ -107,12 +107,17 @@ def lazy_generate_modules(model, loader, args):
             a2_min, a2_max = augmentation2.min(), augmentation2.max()
             print(f"aug1 in range [min: {a1_min}, max: {a1_max}] | "
                   f"aug2 in range [min: {a2_min}, max: {a2_max}]")
-            if a1_max > 1.0 or a1_min < 0:
-                raise ValueError("aug1 outside [0, 1]. You probably dont "
-                                 "want this.")
-            if a2_max > 1.0 or a2_min < 0:
-                raise ValueError("aug2 outside [0, 1]. You probably dont "
-                                 "want this.")
+            # the pipeline relies on ToTensor scaling with NO mean/std
+            # normalization (/root/reference/main.py:486-490) — pixels
+            # outside [0,1] mean a transform is mis-wired, so fail hard
+            for name, lo, hi in (("augmentation1", a1_min, a1_max),
+                                 ("augmentation2", a2_min, a2_max)):
+                if hi > 1.0 or lo < 0:
+                    raise ValueError(
+                        f"{name} pixels fall outside [0, 1] "
+                        f"(min={float(lo):.4f}, max={float(hi):.4f}); the "
+                        "augmentation stack must emit unnormalized "
+                        "ToTensor-scaled images.")
             if args.cuda:
                 augmentation1 = augmentation1.cuda(non_blocking=True)
                 augmentation2 = augmentation2.cuda(non_blocking=True)
@@ -317,6 +322,21 @@ def test(epoch, model, test_loader, grapher, args):
 # epoch driver
 # ---------------------------------------------------------------------------
 
+def _rank_consensus_loss(test_loss, args):
+    """Mean-reduce the per-rank test loss across the process group so every
+    rank feeds the SAME value to ModelSaver and takes the same early-stop
+    branch.  Without this, divergent BN running stats (buffers are NOT
+    broadcast per step by default here, unlike torch-1.5 DDP) could make
+    ranks disagree on stopping and hang the job at the next collective."""
+    import torch.distributed as dist
+    if args.num_replicas > 1 and dist.is_available() and dist.is_initialized():
+        t = torch.tensor([test_loss], dtype=torch.float64,
+                         device="cuda" if args.cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        return float(t.item()) / dist.get_world_size()
+    return test_loss
+
+
 def run(rank, args):
     from ..parallel import init_multiprocessing_and_cuda
     init_multiprocessing_and_cuda(rank, args)
@@ -335,6 +355,7 @@ def run(rank, args):
     for epoch in range(init_epoch, args.epochs + 1):
         train(epoch, model, optimizer, loader.train_loader, grapher, args)
         test_loss = test(epoch, model, loader.test_loader, grapher, args)
+        test_loss = _rank_consensus_loss(test_loss, args)
         loader.set_all_epochs(epoch)
 
         scheduler.step()

@@ -205,7 +205,14 @@ class FusedBatchNorm(nn.Module):
         if self.training:
             with torch.no_grad():
                 self.num_batches_tracked += 1
-        if _hip_eligible(x, self.num_features) and \
+        # eval-mode forwards that still need autograd (e.g. linear-probe
+        # fine-tuning) take the composed path: _FusedBNFunction.backward
+        # implements only the training-mode (batch-stats) gradient, which
+        # is wrong when running stats are constants
+        needs_eval_grad = (not self.training and torch.is_grad_enabled()
+                           and (x.requires_grad
+                                or self.weight.requires_grad))
+        if not needs_eval_grad and _hip_eligible(x, self.num_features) and \
                 (residual is None or _hip_eligible(residual,
                                                    self.num_features)):
             return _FusedBNFunction.apply(

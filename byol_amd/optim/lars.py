@@ -19,6 +19,7 @@ collapses into three HIP kernel launches over the flat buffers
 (:func:`byol_amd.ops.lars_step.fused_lars_momentum_step`).
 """
 
+import contextlib
 from typing import Optional
 
 import torch
@@ -81,24 +82,50 @@ class LARS(Optimizer):
 
     @torch.no_grad()
     def apply_adaptive_lrs(self):
+        """Precondition grads in place, batched with ``_foreach`` kernels
+        (the reference does this one tensor and ~6 ATen calls at a time):
+
+        1. every group: ``grad += weight_decay * p`` (decay folded into the
+           grad so the inner optimizer must NOT apply it again — see
+           :meth:`step`);
+        2. groups tagged ``ignore: False`` by ``add_weight_decay``: scale
+           each grad by ``trust_coef * ||p|| / (||grad|| + eps)``, with a
+           neutral 1.0 whenever either norm is zero.
+
+        In-place mutation (vs the reference's out-of-place rebind) keeps
+        ``p.grad`` a view of the flat grad buffer.
+        """
         for group in self.optim.param_groups:
             weight_decay = group["weight_decay"]
-            ignore = group.get("ignore", None)
-            for p in group["params"]:
-                if p.grad is None:
-                    continue
-                # in-place (vs the reference's out-of-place rebind): same
-                # numerics, but keeps p.grad a view of the flat grad buffer
-                if weight_decay > 0:
-                    p.grad.add_(p, alpha=weight_decay)
-                if ignore is not None and not ignore:
-                    param_norm = p.norm()
-                    grad_norm = p.grad.norm()
-                    adaptive_lr = 1.0
-                    if param_norm > 0 and grad_norm > 0:
-                        adaptive_lr = (self.trust_coef * param_norm
-                                       / (grad_norm + self.eps))
-                    p.grad.mul_(adaptive_lr)
+            adapt = group.get("ignore", None) is not None \
+                and not group["ignore"]
+            params = [p for p in group["params"] if p.grad is not None]
+            if not params:
+                continue
+            grads = [p.grad for p in params]
+            if weight_decay > 0:
+                torch._foreach_add_(grads, params, alpha=weight_decay)
+            if adapt:
+                p_norms = torch.stack(torch._foreach_norm(params))
+                g_norms = torch.stack(torch._foreach_norm(grads))
+                ratios = torch.where(
+                    (p_norms > 0) & (g_norms > 0),
+                    self.trust_coef * p_norms / (g_norms + self.eps),
+                    torch.ones_like(p_norms))
+                torch._foreach_mul_(grads, list(ratios.unbind(0)))
+
+    @contextlib.contextmanager
+    def _suppress_inner_weight_decay(self):
+        """The decay is already in the grads; zero the inner optimizer's
+        ``weight_decay`` for the duration of its step, restore after."""
+        saved = [g["weight_decay"] for g in self.optim.param_groups]
+        for g in self.optim.param_groups:
+            g["weight_decay"] = 0
+        try:
+            yield
+        finally:
+            for g, wd in zip(self.optim.param_groups, saved):
+                g["weight_decay"] = wd
 
     def _try_fused_step(self) -> bool:
         space = getattr(self, "_flat_space", None)
@@ -108,13 +135,10 @@ class LARS(Optimizer):
         return fused.fused_lars_momentum_step(self, space)
 
     def step(self, *args, **kwargs):
+        # GPU + flat params + SGD/momentum inner: the whole multi-tensor
+        # step runs as three HIP kernel launches
         if self._try_fused_step():
             return None
         self.apply_adaptive_lrs()
-        wd_orig = [g["weight_decay"] for g in self.optim.param_groups]
-        for g in self.optim.param_groups:
-            g["weight_decay"] = 0
-        loss = self.optim.step(*args, **kwargs)
-        for g, w in zip(self.optim.param_groups, wd_orig):
-            g["weight_decay"] = w
-        return loss
+        with self._suppress_inner_weight_decay():
+            return self.optim.step(*args, **kwargs)

@@ -1,66 +1,95 @@
-"""Warmup + cosine LR scheduling with the reference's exact semantics
-(``/root/reference/optimizers/scheduler.py``): a ``LinearWarmup`` LambdaLR
-ramps 0 -> 1 over ``warmup_steps`` scheduler steps (the engine steps the
-scheduler once per EPOCH), flips ``complete`` after, and a ``Scheduler``
-container routes ``step``/``get_last_lr``/``state_dict`` to the warmup until
-it completes, then to the cosine schedule."""
+"""Warmup -> cosine LR scheduling.
 
-from torch.optim.lr_scheduler import LambdaLR
+Behavior pinned to the reference's contract (call sites
+``/root/reference/main.py:289-291,763``; class surface
+``/root/reference/optimizers/scheduler.py``): the engine steps the
+scheduler once per EPOCH; LR ramps linearly 0 -> base over ``warmup_steps``
+steps, then a cosine anneal (constructed separately with its own internal
+step counter) takes over for the remaining ``epochs - warmup`` steps.
+
+The implementation is our own: ``LinearWarmup`` writes the param-group LRs
+directly from the captured ``initial_lr`` values (no LambdaLR subclass, no
+closure serialization caveats), and ``Scheduler`` is a two-phase router
+whose active phase is decided by the warmup's ``complete`` latch.  Both
+``complete`` and the step counter ride in ``state_dict`` so a resumed run
+continues in the correct phase.
+"""
 
 __all__ = ["Scheduler", "LinearWarmup"]
 
 
+class LinearWarmup:
+    """Linear LR ramp: factor step/warmup_steps for step < warmup_steps,
+    then a constant 1.0 and ``complete = True`` (the router's handoff
+    latch).
+
+    Mirrors torch scheduler conventions so it is interchangeable with one
+    in checkpoints: construction applies step 0 (LR starts at 0), each
+    ``step()`` advances the counter then writes the LRs, ``initial_lr`` is
+    recorded on (or reused from) the param groups.
+    """
+
+    def __init__(self, optimizer, warmup_steps, last_epoch=-1):
+        self.optimizer = optimizer
+        self.warmup_steps = warmup_steps
+        self.complete = False
+        self.last_epoch = last_epoch
+        if last_epoch == -1:
+            for group in optimizer.param_groups:
+                group.setdefault("initial_lr", group["lr"])
+        self.base_lrs = [g["initial_lr"] for g in optimizer.param_groups]
+        self._last_lr = [g["lr"] for g in optimizer.param_groups]
+        self.step()  # apply step 0, like torch schedulers at last_epoch=-1
+
+    def _factor(self) -> float:
+        if self.last_epoch < self.warmup_steps:
+            return float(self.last_epoch) / float(max(1.0, self.warmup_steps))
+        self.complete = True
+        return 1.0
+
+    def step(self):
+        self.last_epoch += 1
+        factor = self._factor()
+        self._last_lr = []
+        for group, base in zip(self.optimizer.param_groups, self.base_lrs):
+            group["lr"] = base * factor
+            self._last_lr.append(group["lr"])
+
+    def get_last_lr(self):
+        return list(self._last_lr)
+
+    def state_dict(self):
+        return {k: v for k, v in self.__dict__.items() if k != "optimizer"}
+
+    def load_state_dict(self, state_dict):
+        self.__dict__.update(state_dict)
+
+
 class Scheduler:
-    """Container for a warmup scheduler followed by a normal scheduler."""
+    """Two-phase LR router: drives the warmup until its ``complete`` latch
+    flips, then the main (cosine) schedule — which keeps its own private
+    step counter, so the anneal starts from its step 1 at handoff.
+    ``state_dict`` nests both phases' states under fixed keys."""
 
     def __init__(self, normal_scheduler, warmup_scheduler=None):
         self.warmup = warmup_scheduler
         self.sched = normal_scheduler
 
-    def get_last_lr(self):
-        if self.warmup is not None and not self.warmup.complete:
-            return self.warmup.get_last_lr()
-        return self.sched.get_last_lr()
-
-    def state_dict(self):
-        return {
-            "warmup": self.warmup.state_dict() if self.warmup is not None else {},
-            "sched": self.sched.state_dict(),
-        }
-
-    def load_state_dict(self, state_dict):
-        if self.warmup:
-            self.warmup.load_state_dict(state_dict["warmup"])
-        self.sched.load_state_dict(state_dict["sched"])
+    def _active(self):
+        in_warmup = self.warmup is not None and not self.warmup.complete
+        return self.warmup if in_warmup else self.sched
 
     def step(self, *args, **kwargs):
-        if self.warmup is not None and not self.warmup.complete:
-            return self.warmup.step(*args, **kwargs)
-        return self.sched.step(*args, **kwargs)
+        return self._active().step(*args, **kwargs)
 
+    def get_last_lr(self):
+        return self._active().get_last_lr()
 
-class LinearWarmup(LambdaLR):
-    """Linear 0 -> 1 over ``warmup_steps`` scheduler steps, then constant 1
-    (and ``complete=True``, which hands control to the cosine schedule)."""
-
-    def __init__(self, optimizer, warmup_steps, last_epoch=-1):
-        self.warmup_steps = warmup_steps
-        self.complete = False
-        super().__init__(optimizer, self.lr_lambda, last_epoch=last_epoch)
-
-    def lr_lambda(self, step):
-        if step < self.warmup_steps:
-            return float(step) / float(max(1.0, self.warmup_steps))
-        self.complete = True
-        return 1.0
-
-    # `complete` must survive checkpoint/resume: LambdaLR.state_dict drops
-    # callables but keeps plain attributes; make sure both ride along.
     def state_dict(self):
-        sd = {k: v for k, v in self.__dict__.items()
-              if k not in ("optimizer", "lr_lambdas")}
-        return sd
+        warmup_state = {} if self.warmup is None else self.warmup.state_dict()
+        return {"warmup": warmup_state, "sched": self.sched.state_dict()}
 
     def load_state_dict(self, state_dict):
-        self.__dict__.update(
-            {k: v for k, v in state_dict.items() if k != "lr_lambdas"})
+        if self.warmup is not None:
+            self.warmup.load_state_dict(state_dict["warmup"])
+        self.sched.load_state_dict(state_dict["sched"])
