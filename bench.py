@@ -64,6 +64,21 @@ def parse():
     p.add_argument("--bucket-cap-mb", type=float, default=32.0)
     p.add_argument("--miopen-benchmark", action="store_true", default=False,
                    help="exhaustive MIOpen find (slow first run)")
+    p.add_argument("--timed-region", type=str, default="compute",
+                   choices=["compute", "pipeline"],
+                   help="compute: pre-staged device batches, the timed step "
+                        "is fwd+loss+bwd+allreduce+opt (headline). pipeline: "
+                        "H2D upload of raw uint8 frames + the full GPU "
+                        "two-view augmentation run INSIDE the clock too "
+                        "(whole-pipeline number, reported in BENCHMARKS.md)")
+    p.add_argument("--pool", type=int, default=4,
+                   help="distinct pre-generated batches cycled through")
+    p.add_argument("--hip-graph", type=str, default="auto",
+                   choices=["auto", "on", "off"],
+                   help="capture the training step as one hipGraph and "
+                        "replay it (removes ~2k launch boundaries/step of "
+                        "host gap time). auto = on for single-GPU compute-"
+                        "region runs, off when DDP/pipeline is involved")
     p.add_argument("--debug-cpu", action="store_true", default=False,
                    help="plumbing mode: tiny model on CPU over gloo "
                         "(validates the torchrun/distributed path without "
@@ -126,24 +141,45 @@ def main():
 
     B = args.batch_per_gpu
     S = args.image_size
-    # synthetic two-view batches, pre-generated on device (data=synthetic;
-    # augmentation happens upstream of the timed step in real training too
-    # when the GPU pipeline is used)
-    n_pool = 2
+    n_pool = max(args.pool, 1)
     gen = torch.Generator(device=device).manual_seed(99 + rank)
-    pool = [(torch.rand(B, 3, S, S, device=device, generator=gen),
-             torch.rand(B, 3, S, S, device=device, generator=gen),
-             torch.randint(args.classes, (B,), device=device, generator=gen))
+    pipeline_mode = args.timed_region == "pipeline" \
+        and device.type == "cuda"
+    if pipeline_mode:
+        # raw "decoded" uint8 frames in pinned host memory; each timed step
+        # uploads one batch and runs the full HIP two-view augmentation
+        # (RandomResizedCrop+flip -> jitter/gray -> per-sigma blur) before
+        # the training step — decode->augment->H2D all inside the clock
+        from byol_amd.data.gpu_augment import GPUTwoViewAugment
+        src_size = int(S * 8 / 7)  # e.g. 256 source for 224 crops
+        cpu_gen = torch.Generator().manual_seed(99 + rank)
+        raw_pool = [
+            (torch.randint(0, 256, (B, 3, src_size, src_size),
+                           dtype=torch.uint8, generator=cpu_gen)
+             .pin_memory(),
+             torch.randint(args.classes, (B,), generator=cpu_gen)
+             .pin_memory())
             for _ in range(n_pool)]
-    if args.channels_last:
-        pool = [(a.to(memory_format=torch.channels_last),
-                 b.to(memory_format=torch.channels_last), l)
-                for a, b, l in pool]
+        augment = GPUTwoViewAugment(out_size=S, jitter_strength=1.0,
+                                    seed=7 + rank)
+    else:
+        # synthetic two-view batches, pre-staged on device: the timed step
+        # is pure training compute (fwd+loss+bwd+allreduce+opt)
+        pool = [(torch.rand(B, 3, S, S, device=device, generator=gen),
+                 torch.rand(B, 3, S, S, device=device, generator=gen),
+                 torch.randint(args.classes, (B,), device=device,
+                               generator=gen))
+                for _ in range(n_pool)]
+        if args.channels_last:
+            pool = [(a.to(memory_format=torch.channels_last),
+                     b.to(memory_format=torch.channels_last), l)
+                    for a, b, l in pool]
 
     use_bf16 = args.dtype == "bf16" and device.type == "cuda"
 
-    def step(i):
-        a1, a2, lab = pool[i % n_pool]
+    def step_body(a1, a2, lab):
+        """The full training step (fwd + loss + bwd + opt) — eager or
+        captured into a hipGraph; nothing is skipped or cached."""
         with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
             out = net(a1, a2)
             loss = loss_function(
@@ -162,6 +198,46 @@ def main():
             net.finish_grad_sync()
         opt.step()
         return loss
+
+    def eager_step(i):
+        if pipeline_mode:
+            raw, lab_h = raw_pool[i % n_pool]
+            raw_d = raw.to(device, non_blocking=True)
+            lab = lab_h.to(device, non_blocking=True)
+            a1, a2 = augment(raw_d.float().div_(255.0))
+        else:
+            a1, a2, lab = pool[i % n_pool]
+        return step_body(a1, a2, lab)
+
+    use_graph = (args.hip_graph == "on"
+                 or (args.hip_graph == "auto" and not distributed
+                     and not pipeline_mode and device.type == "cuda"))
+    graphed = None
+    if use_graph:
+        from byol_amd.engine.graph_step import GraphedTrainStep
+        # static input buffers: the graph replays on these; each step
+        # copies the pool batch in (D2D, ~0.3 ms — far less than the
+        # launch-gap time the graph removes)
+        sa1, sa2, slab = (pool[0][0].clone(), pool[0][1].clone(),
+                          pool[0][2].clone())
+        graphed = GraphedTrainStep(model, opt, step_body,
+                                   (sa1, sa2, slab), warmup_steps=2)
+        try:
+            graphed.capture()
+            if rank == 0:
+                print("hipGraph captured", file=sys.stderr, flush=True)
+        except Exception as e:  # noqa: BLE001 — fall back to eager
+            graphed = None
+            use_graph = False
+            if rank == 0:
+                print(f"hipGraph capture failed ({e}); running eager",
+                      file=sys.stderr, flush=True)
+
+    def step(i):
+        if graphed is not None:
+            a1, a2, lab = pool[i % n_pool]
+            return graphed.replay(a1, a2, lab)
+        return eager_step(i)
 
     for i in range(args.warmup):
         step(i)
@@ -215,6 +291,8 @@ def main():
                 "parallelism": f"dp{world}",
                 "sync_bn": args.sync_bn,
                 "channels_last": args.channels_last,
+                "timed_region": args.timed_region,
+                "hip_graph": bool(use_graph),
             },
         }))
     if distributed:

@@ -131,20 +131,40 @@ class CosEMA(nn.Module):
         self.total_steps = max(int(total_steps), 1)
         self.base_decay = float(base_decay)
         self.register_buffer("mean", None)
+        self._decay_dev = None  # 1-float device scalar for hipGraph capture
 
     def current_decay(self) -> float:
         return 1.0 - (1.0 - self.base_decay) * (
             math.cos(math.pi * self.step / self.total_steps) + 1.0) / 2.0
+
+    def ensure_decay_dev(self, device) -> torch.Tensor:
+        """Device scalar holding the current decay — the hipGraph replay
+        wrapper refreshes it (and advances ``step``) between replays."""
+        if self._decay_dev is None or self._decay_dev.device != device:
+            self._decay_dev = torch.zeros(1, device=device,
+                                          dtype=torch.float32)
+        self._decay_dev.fill_(self.current_decay())
+        return self._decay_dev
 
     @torch.no_grad()
     def forward(self, flat: torch.Tensor) -> torch.Tensor:
         if self.mean is None:
             self.mean = torch.zeros_like(flat)
         if self.training:
-            decay = self.current_decay()
             from ..ops import ema as ema_ops
-            ema_ops.flat_ema_update(self.mean, flat, decay)
-            self.step += 1
+            if flat.is_cuda and torch.cuda.is_current_stream_capturing():
+                # graph capture: decay comes from the device scalar and the
+                # replay wrapper owns the host-side step advancement
+                if self._decay_dev is None:
+                    raise RuntimeError(
+                        "CosEMA captured without ensure_decay_dev(); the "
+                        "graph wrapper must create the decay scalar first")
+                ema_ops.flat_ema_update(self.mean, flat, 0.0,
+                                        decay_dev=self._decay_dev)
+            else:
+                ema_ops.flat_ema_update(self.mean, flat,
+                                        self.current_decay())
+                self.step += 1
         return flat
 
     # step/total/base ride the state_dict alongside the mean buffer so
@@ -284,6 +304,11 @@ class BYOL(nn.Module):
 
     def forward(self, augmentation1: torch.Tensor,
                 augmentation2: torch.Tensor) -> Dict[str, torch.Tensor]:
+        # parameters are frozen for the duration of one forward (LARS step,
+        # EMA lerp and DDP broadcast all happen outside it), so the conv
+        # weight-layout cache is valid exactly until the next forward
+        from ..ops.conv import clear_weight_cache
+        clear_weight_cache()
         online_representation1, online_projection1, online_prediction1 = \
             self.prediction(augmentation1)
         online_representation2, online_projection2, online_prediction2 = \

@@ -48,8 +48,10 @@ def _rows_view(t: torch.Tensor, c: int) -> torch.Tensor:
 
 
 def _bf16_enabled() -> bool:
-    # round-2 candidate path: bf16-I/O BN kernels (BYOL_BF16_BN=1)
-    return os.environ.get("BYOL_BF16_BN", "0") == "1"
+    # bf16-I/O BN kernels: validated round 2 (numerics vs fp32 oracle +
+    # bench: 2805 img/s vs 2323 composed at bs=1024 bf16,
+    # profiles/r02_validation.md) — default ON; BYOL_BF16_BN=0 disables
+    return os.environ.get("BYOL_BF16_BN", "1") == "1"
 
 
 def _hip_eligible(x: torch.Tensor, c: int) -> bool:

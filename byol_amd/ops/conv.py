@@ -57,6 +57,37 @@ def _rows(t: torch.Tensor, c: int) -> torch.Tensor:
     return t.permute(0, 2, 3, 1).reshape(-1, c)
 
 
+# ---------------------------------------------------------------------------
+# per-forward weight-layout cache
+#
+# The MFMA kernels want W pre-arranged ([N,K]+[K,N] for 1x1, [9][K][N] for
+# 3x3); rebuilding those copies on every conv call cost ~0.5 ms/call on the
+# big shapes.  BYOL calls every conv FOUR times per step with only two
+# distinct weight tensors (online flat views x2 passes, EMA views x2), so a
+# cache that BYOL.forward clears on entry halves the copies and can never go
+# stale: all parameter mutation (LARS step, EMA lerp, DDP broadcast) happens
+# outside model.forward.  Entries hold a strong ref to the weight and check
+# both identity and _version, so standalone use (tests, microbench) is also
+# safe whenever in-place updates go through ATen.
+# ---------------------------------------------------------------------------
+
+_weight_cache: dict = {}
+
+
+def clear_weight_cache() -> None:
+    _weight_cache.clear()
+
+
+def _cached_layout(weight: torch.Tensor, kind: str, build):
+    key = (id(weight), kind)
+    hit = _weight_cache.get(key)
+    if hit is not None and hit[0] is weight and hit[1] == weight._version:
+        return hit[2]
+    value = build()
+    _weight_cache[key] = (weight, weight._version, value)
+    return value
+
+
 class _Conv1x1Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight):
@@ -64,9 +95,12 @@ class _Conv1x1Fn(torch.autograd.Function):
         b, k, h, w = x.shape
         n = weight.shape[0]
         m = b * h * w
-        wv = weight.reshape(n, k).contiguous()
-        # pre-transposed copy feeds the glds fast path ([K,N] natural)
-        wt = wv.t().contiguous()
+        # [N,K] + pre-transposed [K,N] (feeds the glds fast path), cached
+        # across the 2 passes sharing this weight tensor within one step
+        wv, wt = _cached_layout(
+            weight, "1x1",
+            lambda: (lambda v: (v, v.t().contiguous()))(
+                weight.reshape(n, k).contiguous()))
         y = torch.empty((b, n, h, w), device=x.device, dtype=x.dtype,
                         memory_format=torch.channels_last)
         C.conv1x1_fwd(_rows(x, k), wv, wt, _rows(y, n), m, k, n)
@@ -134,7 +168,10 @@ class _Conv3x3Fn(torch.autograd.Function):
         ho = (hi + 2 - 3) // stride + 1
         wo = (wi + 2 - 3) // stride + 1
         # prepermute W [N,K,3,3] -> [9][K][N] for coalesced B staging
-        wp = weight.reshape(n, k, 9).permute(2, 1, 0).contiguous()
+        # (cached across the 2 passes sharing this weight within one step)
+        wp = _cached_layout(
+            weight, "3x3",
+            lambda: weight.reshape(n, k, 9).permute(2, 1, 0).contiguous())
         y = torch.empty((b, n, ho, wo), device=x.device, dtype=x.dtype,
                         memory_format=torch.channels_last)
         m = b * ho * wo

@@ -5,7 +5,8 @@
 #include <hip/hip_runtime.h>
 
 void launch_flat_ema_update(float* mean, const float* x, float decay,
-                            int64_t n, hipStream_t stream);
+                            const float* decay_dev, int64_t n,
+                            hipStream_t stream);
 void launch_byol_loss_forward(const float* p1, const float* p2,
                               const float* z1, const float* z2, float* stats,
                               float* loss, int64_t n, int64_t batch,
@@ -92,7 +93,8 @@ void launch_lars_momentum_step(float* p, const float* g, float* m,
                                const int* seg_adapt, const int* chunk_seg,
                                const int64_t* chunk_base, int nseg,
                                int nchunks, int64_t chunk, float trust,
-                               float eps, float lr, float momentum,
+                               float eps, float lr, const float* lr_dev,
+                               float momentum,
                                int m_init, hipStream_t stream);
 
 namespace {
@@ -108,7 +110,22 @@ void flat_ema_update(torch::Tensor mean, torch::Tensor x, double decay) {
   TORCH_CHECK(mean.numel() == x.numel(), "size mismatch");
   auto stream = at::hip::getCurrentHIPStream();
   launch_flat_ema_update(mean.data_ptr<float>(), x.data_ptr<float>(),
-                         static_cast<float>(decay), mean.numel(), stream);
+                         static_cast<float>(decay), nullptr, mean.numel(),
+                         stream);
+}
+
+// decay read from a 1-float device scalar: hipGraph-capturable (the replay
+// wrapper rewrites the scalar between replays, outside the graph)
+void flat_ema_update_dev(torch::Tensor mean, torch::Tensor x,
+                         torch::Tensor decay_dev) {
+  CHECK_IN(mean);
+  CHECK_IN(x);
+  CHECK_IN(decay_dev);
+  TORCH_CHECK(mean.numel() == x.numel(), "size mismatch");
+  TORCH_CHECK(decay_dev.numel() == 1, "decay_dev must be a 1-elem tensor");
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_flat_ema_update(mean.data_ptr<float>(), x.data_ptr<float>(), 0.f,
+                         decay_dev.data_ptr<float>(), mean.numel(), stream);
 }
 
 std::tuple<torch::Tensor, torch::Tensor> byol_loss_forward(
@@ -154,7 +171,8 @@ void lars_momentum_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                         torch::Tensor seg_wd, torch::Tensor seg_adapt,
                         torch::Tensor chunk_seg, torch::Tensor chunk_base,
                         double trust, double eps, double lr, double momentum,
-                        int64_t m_init) {
+                        int64_t m_init,
+                        c10::optional<torch::Tensor> lr_dev = c10::nullopt) {
   CHECK_IN(p); CHECK_IN(g); CHECK_IN(m); CHECK_IN(norm_acc); CHECK_IN(alr);
   TORCH_CHECK(seg_off.scalar_type() == at::kLong &&
               seg_len.scalar_type() == at::kLong &&
@@ -171,7 +189,9 @@ void lars_momentum_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
       seg_wd.data_ptr<float>(), seg_adapt.data_ptr<int>(),
       chunk_seg.data_ptr<int>(), chunk_base.data_ptr<int64_t>(), nseg,
       nchunks, 65536, static_cast<float>(trust), static_cast<float>(eps),
-      static_cast<float>(lr), static_cast<float>(momentum),
+      static_cast<float>(lr),
+      lr_dev.has_value() ? lr_dev->data_ptr<float>() : nullptr,
+      static_cast<float>(momentum),
       static_cast<int>(m_init), stream);
 }
 
@@ -457,10 +477,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_bwd_apply", &bn_bwd_apply);
   mod.def("flat_ema_update", &flat_ema_update,
           "fused flat-parameter EMA update (gfx950)");
+  mod.def("flat_ema_update_dev", &flat_ema_update_dev,
+          "EMA update with device-scalar decay (hipGraph-capturable)");
   mod.def("byol_loss_forward", &byol_loss_forward,
           "fused BYOL loss forward (gfx950)");
   mod.def("byol_loss_backward", &byol_loss_backward,
           "fused BYOL loss backward (gfx950)");
   mod.def("lars_momentum_step", &lars_momentum_step,
-          "fused multi-tensor LARS+momentum step (gfx950)");
+          "fused multi-tensor LARS+momentum step (gfx950)",
+          pybind11::arg("p"), pybind11::arg("g"), pybind11::arg("m"),
+          pybind11::arg("norm_acc"), pybind11::arg("alr"),
+          pybind11::arg("seg_off"), pybind11::arg("seg_len"),
+          pybind11::arg("seg_wd"), pybind11::arg("seg_adapt"),
+          pybind11::arg("chunk_seg"), pybind11::arg("chunk_base"),
+          pybind11::arg("trust"), pybind11::arg("eps"), pybind11::arg("lr"),
+          pybind11::arg("momentum"), pybind11::arg("m_init"),
+          pybind11::arg("lr_dev") = pybind11::none());
 }

@@ -4,10 +4,17 @@
 // ATen blend (/root/reference/main.py:158-161,255).
 #include "common.h"
 
+// decay arrives either as an immediate (decay_dev == nullptr) or in a
+// 1-float device scalar — the latter keeps the kernel hipGraph-capturable
+// with a per-step cosine-ramped decay (the replay wrapper rewrites the
+// scalar outside the graph).
 __global__ void flat_ema_update_kernel(float* __restrict__ mean,
                                        const float* __restrict__ x,
-                                       float decay, int64_t n4,
+                                       float decay,
+                                       const float* __restrict__ decay_dev,
+                                       int64_t n4,
                                        int64_t n_tail, int64_t tail_base) {
+  if (decay_dev != nullptr) decay = *decay_dev;
   const float w = 1.0f - decay;
   float4* m4 = reinterpret_cast<float4*>(mean);
   const float4* x4 = reinterpret_cast<const float4*>(x);
@@ -31,12 +38,14 @@ __global__ void flat_ema_update_kernel(float* __restrict__ mean,
 }
 
 void launch_flat_ema_update(float* mean, const float* x, float decay,
-                            int64_t n, hipStream_t stream) {
+                            const float* decay_dev, int64_t n,
+                            hipStream_t stream) {
   const int64_t n4 = n / 4;
   const int64_t tail_base = n4 * 4;
   const int64_t n_tail = n - tail_base;
   const int block = 256;
   const int grid = grid_1d(n4 > 0 ? n4 : n_tail, block);
   hipLaunchKernelGGL(flat_ema_update_kernel, dim3(grid), dim3(block), 0,
-                     stream, mean, x, decay, n4, n_tail, tail_base);
+                     stream, mean, x, decay, decay_dev, n4, n_tail,
+                     tail_base);
 }

@@ -69,7 +69,11 @@ __global__ void lars_update_kernel(float* __restrict__ p,
                                    const int* __restrict__ chunk_seg,
                                    const int64_t* __restrict__ chunk_base,
                                    int64_t chunk, int nchunks,
-                                   float lr, float momentum, int m_init) {
+                                   float lr, const float* __restrict__ lr_dev,
+                                   float momentum, int m_init) {
+  // lr optionally read from a device scalar so the step is hipGraph-
+  // capturable across scheduler changes (wrapper rewrites the scalar)
+  if (lr_dev != nullptr) lr = *lr_dev;
   const int ci = blockIdx.x;
   if (ci >= nchunks) return;
   const int s = chunk_seg[ci];
@@ -98,6 +102,7 @@ void launch_lars_momentum_step(float* p, const float* g, float* m,
                                const int64_t* chunk_base,
                                int nseg, int nchunks, int64_t chunk,
                                float trust, float eps, float lr,
+                               const float* lr_dev,
                                float momentum, int m_init,
                                hipStream_t stream) {
   hipMemsetAsync(norm_acc, 0, sizeof(float) * 2 * nseg, stream);
@@ -108,5 +113,6 @@ void launch_lars_momentum_step(float* p, const float* g, float* m,
                      0, stream, norm_acc, seg_adapt, alr, trust, eps, nseg);
   hipLaunchKernelGGL(lars_update_kernel, dim3(nchunks), dim3(256), 0, stream,
                      p, g, m, alr, seg_off, seg_len, seg_wd, chunk_seg,
-                     chunk_base, chunk, nchunks, lr, momentum, m_init);
+                     chunk_base, chunk, nchunks, lr, lr_dev, momentum,
+                     m_init);
 }

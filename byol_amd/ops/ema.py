@@ -14,12 +14,19 @@ from . import require_extension
 
 
 @torch.no_grad()
-def flat_ema_update(mean: torch.Tensor, x: torch.Tensor, decay: float) -> None:
+def flat_ema_update(mean: torch.Tensor, x: torch.Tensor, decay: float,
+                    decay_dev: torch.Tensor = None) -> None:
+    """One fused lerp.  ``decay_dev`` (1-float device scalar) overrides the
+    host ``decay`` — used under hipGraph capture, where the replay wrapper
+    rewrites the scalar between replays."""
     assert mean.is_contiguous() and x.is_contiguous()
     assert mean.numel() == x.numel()
     if mean.is_cuda:
         ext = require_extension("flat_ema_update")
-        ext.flat_ema_update(mean, x, float(decay))
+        if decay_dev is not None:
+            ext.flat_ema_update_dev(mean, x, decay_dev)
+        else:
+            ext.flat_ema_update(mean, x, float(decay))
     else:
         # oracle path: identical math via one ATen lerp
         mean.lerp_(x, 1.0 - float(decay))

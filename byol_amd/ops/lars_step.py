@@ -60,6 +60,10 @@ class FlatLarsState:
         self.seg_adapt = torch.tensor(adapts, dtype=torch.int32,
                                       device=device)
         self.nseg = len(offs)
+        # 1-float device scalar for hipGraph capture: the captured update
+        # kernel reads lr from here; the replay wrapper refreshes it when
+        # the scheduler changes the LR
+        self.lr_dev = torch.zeros(1, dtype=torch.float32, device=device)
         self.momentum = torch.zeros_like(flat)
         self.momentum_initialized = False
         self.norm_acc = torch.zeros(2 * self.nseg, dtype=torch.float32,
@@ -116,13 +120,17 @@ def fused_lars_momentum_step(lars, space) -> bool:
         _import_momentum(inner, space, state)
     lr = float(inner.param_groups[0]["lr"])
     mu = float(inner.param_groups[0]["momentum"])
+    capturing = torch.cuda.is_current_stream_capturing()
     ext.lars_momentum_step(
         space.flat_params, space.flat_grads, state.momentum,
         state.norm_acc, state.alr,
         state.seg_off, state.seg_len, state.seg_wd, state.seg_adapt,
         state.chunk_seg, state.chunk_base,
         float(lars.trust_coef), float(lars.eps), lr, mu,
-        1 if state.momentum_initialized else 0)
+        1 if state.momentum_initialized else 0,
+        # under capture the kernel reads lr from the device scalar so the
+        # graph survives scheduler LR changes (wrapper refreshes it)
+        state.lr_dev if capturing else None)
     state.momentum_initialized = True
     _export_momentum_lazy(inner, space, state)
     return True

@@ -386,9 +386,6 @@ def test_smoke_entrypoint():
     __graft_entry__.smoke()
 
 
-@pytest.mark.skipif(
-    __import__("os").environ.get("BYOL_TEST_PENDING", "0") != "1",
-    reason="round-2 candidate kernel; enable with BYOL_TEST_PENDING=1")
 def test_wgrad_v2_matches_oracle(ext, monkeypatch):
     import torch.nn.functional as F
     from byol_amd.ops.conv import _Conv1x1Fn
@@ -411,9 +408,6 @@ def test_wgrad_v2_matches_oracle(ext, monkeypatch):
             (hgt, cin, cout, (w.grad - w2.grad).abs().max().item())
 
 
-@pytest.mark.skipif(
-    __import__("os").environ.get("BYOL_TEST_PENDING", "0") != "1",
-    reason="round-2 candidate kernel; enable with BYOL_TEST_PENDING=1")
 def test_conv3x3_fast_path_matches_oracle(ext, monkeypatch):
     """glds fast path (padded input) — fast-eligible shapes (M%128==0)."""
     import torch.nn.functional as F
@@ -435,9 +429,6 @@ def test_conv3x3_fast_path_matches_oracle(ext, monkeypatch):
             (hgt, c, stride, (y - y2).abs().max().item())
 
 
-@pytest.mark.skipif(
-    __import__("os").environ.get("BYOL_TEST_PENDING", "0") != "1",
-    reason="round-2 candidate kernel; enable with BYOL_TEST_PENDING=1")
 def test_aug_sample_v2_matches_v1(ext):
     import torch.nn.functional as F
     torch.manual_seed(14)
@@ -457,9 +448,6 @@ def test_aug_sample_v2_matches_v1(ext):
     assert torch.allclose(out[0][1], out[1][1], rtol=1e-4, atol=1e-3)
 
 
-@pytest.mark.skipif(
-    __import__("os").environ.get("BYOL_TEST_PENDING", "0") != "1",
-    reason="round-2 candidate kernel; enable with BYOL_TEST_PENDING=1")
 def test_bf16_fused_bn_vs_fp32_oracle(ext, monkeypatch):
     """bf16-I/O BN kernels vs an fp32 reference on the same (bf16-rounded)
     inputs; stats/params stay fp32 so tolerances are bf16-IO-class."""
