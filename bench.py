@@ -209,9 +209,9 @@ def main():
             a1, a2, lab = pool[i % n_pool]
         return step_body(a1, a2, lab)
 
-    use_graph = (args.hip_graph == "on"
-                 or (args.hip_graph == "auto" and not distributed
-                     and not pipeline_mode and device.type == "cuda"))
+    graph_capable = (device.type == "cuda" and not pipeline_mode
+                     and not distributed)
+    use_graph = graph_capable and args.hip_graph in ("on", "auto")
     graphed = None
     if use_graph:
         from byol_amd.engine.graph_step import GraphedTrainStep

@@ -485,3 +485,117 @@ def test_bf16_fused_bn_vs_fp32_oracle(ext, monkeypatch):
             (x1.grad.float() - x2.grad).abs().max().item()
         assert torch.allclose(fused.running_mean, ref.running_mean,
                               atol=5e-3)
+
+
+def _tiny_byol(seed=33):
+    from byol_amd.models.byol import BYOL
+    torch.manual_seed(seed)
+    m = BYOL(arch="resnet18", base_network_output_size=512,
+             projection_output_size=64, classifier_output_size=10,
+             total_training_steps=100, head_latent_size=128).cuda()
+    m.base_network.to(memory_format=torch.channels_last)
+    m.finalize()
+    m.train()
+    return m
+
+
+def _lars_for(model):
+    from byol_amd import layers
+    from byol_amd.optim.lars import LARS
+    inner = torch.optim.SGD(layers.add_weight_decay(model, 1e-6), lr=0.05,
+                            momentum=0.9)
+    opt = LARS(inner, eps=0.0)
+    opt.attach_flat_space(model.flat_space)
+    return opt
+
+
+def test_hipgraph_step_parity(ext):
+    """Captured-graph replays produce the same parameters as eager steps:
+    same tiny BYOL, same data sequence, 3 steps each way."""
+    from byol_amd.engine.graph_step import GraphedTrainStep
+    from byol_amd.objective import loss_function
+    from byol_amd.ops.classifier import cross_entropy_topk
+
+    torch.manual_seed(44)
+    B, S = 16, 32
+    batches = [(torch.rand(B, 3, S, S, device="cuda")
+                .to(memory_format=torch.channels_last),
+                torch.rand(B, 3, S, S, device="cuda")
+                .to(memory_format=torch.channels_last),
+                torch.randint(10, (B,), device="cuda"))
+               for _ in range(5)]
+
+    def make_step(model, opt):
+        def step_body(a1, a2, lab):
+            out = model(a1, a2)
+            loss = loss_function(
+                out["online_prediction1"].float(),
+                out["online_prediction2"].float(),
+                out["target_projection1"].float(),
+                out["target_projection2"].float())
+            ce, _x, _y = cross_entropy_topk(
+                out["linear_preds"].float().contiguous(),
+                torch.cat([lab, lab]))
+            loss = loss + ce
+            opt.zero_grad()
+            loss.backward()
+            opt.step()
+            return loss
+        return step_body
+
+    # eager: warmup(2) on batches[0] then 3 steps on batches[1..3]
+    m1 = _tiny_byol()
+    o1 = _lars_for(m1)
+    s1 = make_step(m1, o1)
+    for _ in range(2):
+        s1(*batches[0])
+    for i in (1, 2, 3):
+        s1(*batches[i])
+
+    # graphed: identical sequence, steps via replay
+    m2 = _tiny_byol()
+    o2 = _lars_for(m2)
+    s2 = make_step(m2, o2)
+    static = tuple(t.clone() for t in batches[0])
+    g = GraphedTrainStep(m2, o2, s2, static, warmup_steps=2)
+    g.capture()
+    for i in (1, 2, 3):
+        g.replay(*batches[i])
+    torch.cuda.synchronize()
+
+    pa = m1.flat_space.flat_params
+    pb = m2.flat_space.flat_params
+    assert torch.allclose(pa, pb, rtol=1e-4, atol=1e-5), \
+        (pa - pb).abs().max().item()
+    ea = m1.target_network.mean
+    eb = m2.target_network.mean
+    assert m1.target_network.step == m2.target_network.step
+    assert torch.allclose(ea, eb, rtol=1e-4, atol=1e-5), \
+        (ea - eb).abs().max().item()
+
+
+def test_rccl_single_rank_group(ext):
+    """Exercise the real RCCL code path (init + all_reduce + broadcast)
+    with a world-1 process group on the GPU — the multi-rank semantics are
+    gloo-covered (tests/test_ddp_gloo.py); this pins the nccl backend
+    itself working on the box."""
+    import torch.distributed as dist
+    if dist.is_initialized():
+        pytest.skip("process group already initialized")
+    import os
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29571")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        t = torch.arange(1024, device="cuda", dtype=torch.float32)
+        dist.all_reduce(t)
+        dist.broadcast(t, src=0)
+        ref = torch.arange(1024, device="cuda", dtype=torch.float32)
+        assert torch.equal(t, ref)
+        # the packed SyncBN exchange path used by ops/bn.py
+        acc = torch.randn(2 * 64, device="cuda")
+        want = acc.clone()
+        dist.all_reduce(acc)
+        assert torch.allclose(acc, want)
+    finally:
+        dist.destroy_process_group()
