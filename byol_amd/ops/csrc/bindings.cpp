@@ -18,6 +18,8 @@ void launch_byol_loss_backward(const float* p1, const float* p2,
                                int64_t batch, hipStream_t stream);
 void launch_bn_stats(const float* x, float* acc, int64_t m, int c,
                      int slot_mask, hipStream_t stream);
+void launch_bn_stats_v2(const float* x, float* acc, int64_t m, int c,
+                        int slot_mask, int grid, hipStream_t stream);
 void launch_bn_reduce_slots(const float* in, float* out, int n2c, int nslots,
                             hipStream_t stream);
 void launch_bn_finalize(const float* acc, float* mean, float* invstd,
@@ -206,6 +208,15 @@ void bn_stats(torch::Tensor x, torch::Tensor acc, int64_t m, int64_t c,
   auto stream = at::hip::getCurrentHIPStream();
   launch_bn_stats(x.data_ptr<float>(), acc.data_ptr<float>(), m, (int)c,
                   (int)(nslots - 1), stream);
+}
+
+void bn_stats_v2(torch::Tensor x, torch::Tensor acc, int64_t m, int64_t c,
+                 int64_t nslots, int64_t grid) {
+  CHECK_IN(x); CHECK_IN(acc);
+  TORCH_CHECK(acc.numel() == nslots * 2 * c, "acc must be [nslots][2C]");
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_stats_v2(x.data_ptr<float>(), acc.data_ptr<float>(), m, (int)c,
+                     (int)(nslots - 1), (int)grid, stream);
 }
 
 void bn_reduce_slots(torch::Tensor in, torch::Tensor out, int64_t nslots) {
@@ -470,6 +481,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("aug_sample", &aug_sample);
   mod.def("aug_color", &aug_color);
   mod.def("bn_stats", &bn_stats);
+  mod.def("bn_stats_v2", &bn_stats_v2);
   mod.def("bn_reduce_slots", &bn_reduce_slots);
   mod.def("bn_finalize", &bn_finalize);
   mod.def("bn_apply", &bn_apply);

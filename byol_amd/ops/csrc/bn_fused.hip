@@ -442,3 +442,128 @@ void launch_bn_bwd_apply(const float* dy, const float* y, const float* x,
                      0, stream, dy, y, x, mean, invstd, weight, red, dx,
                      dresidual, inv_count, m, c, relu);
 }
+
+// ---------------------------------------------------------------------------
+// stats v2: 8-row unroll + in-wave shuffle tail for small channel widths.
+// Experiment vehicle for the round-2 BN bandwidth track (v1 measured
+// ~3.5 TB/s vs bn_apply's 4.9 — suspicion: not enough loads in flight and
+// a serial per-block reduce tail at small C).  Grid is a launcher argument
+// so the microbench can sweep it.
+// ---------------------------------------------------------------------------
+template <int KQ>
+__global__ void bn_stats_v2_kernel(const float* __restrict__ x,
+                                   float* __restrict__ acc,
+                                   int64_t m, int c, int slot_mask) {
+  float* const acc_slot = acc + (int64_t)(blockIdx.x & slot_mask) * 2 * c;
+  const int c4 = c >> 2;
+  const int nthread = blockDim.x;
+  const int cw = c4 < nthread ? c4 : nthread;
+  const int rpb = nthread / cw;
+  const int tc = threadIdx.x % cw;
+  const int tr = threadIdx.x / cw;
+  const bool active = tr < rpb;
+
+  float4 s[KQ], ss[KQ];
+  #pragma unroll
+  for (int k = 0; k < KQ; ++k) {
+    s[k] = make_float4(0.f, 0.f, 0.f, 0.f);
+    ss[k] = make_float4(0.f, 0.f, 0.f, 0.f);
+  }
+
+  const float4* x4 = reinterpret_cast<const float4*>(x);
+  // 8-row unroll: 8 independent 16-B loads in flight per thread
+  const int64_t row_stride = (int64_t)rpb * 8 * gridDim.x;
+  if (active) {
+    for (int64_t row = (int64_t)blockIdx.x * rpb * 8 + tr; row < m;
+         row += row_stride) {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int64_t r = row + (int64_t)j * rpb;
+        if (r < m) {
+          const int64_t base = r * c4;
+          #pragma unroll
+          for (int k = 0; k < KQ; ++k) {
+            const int q = tc + k * cw;
+            if (q < c4) {
+              const float4 v = x4[base + q];
+              s[k].x += v.x; s[k].y += v.y; s[k].z += v.z; s[k].w += v.w;
+              ss[k] = make_float4(fmaf(v.x, v.x, ss[k].x),
+                                  fmaf(v.y, v.y, ss[k].y),
+                                  fmaf(v.z, v.z, ss[k].z),
+                                  fmaf(v.w, v.w, ss[k].w));
+            }
+          }
+        }
+      }
+    }
+  }
+
+  // tail: reduce across row lanes.  For cw <= 32 (rows share a wave) fold
+  // in-wave first with shfl_down — the v1 serial loop is up to 16 scalar
+  // float4 adds on 1/16 of the threads.
+  __shared__ Quad scratch[256];
+  // shfl tree needs power-of-two cw; otherwise degrade to the v1 serial
+  // tail (rows_in_wave = 1 disables the in-wave fold)
+  const bool cw_pow2 = (cw & (cw - 1)) == 0;
+  const int rows_in_wave =
+      (cw_pow2 && cw < WAVE_SIZE) ? (WAVE_SIZE / cw) : 1;
+  #pragma unroll
+  for (int k = 0; k < KQ; ++k) {
+    float4 ts = s[k], tss = ss[k];
+    for (int off = cw * (rows_in_wave >> 1); off >= cw && off > 0;
+         off >>= 1) {
+      ts.x += __shfl_down(ts.x, off, WAVE_SIZE);
+      ts.y += __shfl_down(ts.y, off, WAVE_SIZE);
+      ts.z += __shfl_down(ts.z, off, WAVE_SIZE);
+      ts.w += __shfl_down(ts.w, off, WAVE_SIZE);
+      tss.x += __shfl_down(tss.x, off, WAVE_SIZE);
+      tss.y += __shfl_down(tss.y, off, WAVE_SIZE);
+      tss.z += __shfl_down(tss.z, off, WAVE_SIZE);
+      tss.w += __shfl_down(tss.w, off, WAVE_SIZE);
+    }
+    // after the in-wave fold, row-lane tr % rows_in_wave == 0 holds the
+    // wave-partial; cross-wave partials go through LDS
+    const bool wave_leader = active && (tr % rows_in_wave) == 0;
+    scratch[threadIdx.x].s = ts;
+    scratch[threadIdx.x].ss = tss;
+    __syncthreads();
+    if (tr == 0) {
+      float4 fs = ts, fss = tss;
+      for (int r = rows_in_wave; r < rpb; r += rows_in_wave) {
+        const Quad& o = scratch[tc + r * cw];
+        fs.x += o.s.x; fs.y += o.s.y; fs.z += o.s.z; fs.w += o.s.w;
+        fss.x += o.ss.x; fss.y += o.ss.y; fss.z += o.ss.z;
+        fss.w += o.ss.w;
+      }
+      const int q = tc + k * cw;
+      if (q < c4) {
+        const int ch = q * 4;
+        atomicAdd(&acc_slot[ch + 0], fs.x);
+        atomicAdd(&acc_slot[ch + 1], fs.y);
+        atomicAdd(&acc_slot[ch + 2], fs.z);
+        atomicAdd(&acc_slot[ch + 3], fs.w);
+        atomicAdd(&acc_slot[c + ch + 0], fss.x);
+        atomicAdd(&acc_slot[c + ch + 1], fss.y);
+        atomicAdd(&acc_slot[c + ch + 2], fss.z);
+        atomicAdd(&acc_slot[c + ch + 3], fss.w);
+      }
+    }
+    __syncthreads();
+    (void)wave_leader;
+  }
+}
+
+void launch_bn_stats_v2(const float* x, float* acc, int64_t m, int c,
+                        int slot_mask, int grid, hipStream_t stream) {
+  const dim3 g(grid), b(256);
+  switch (kq_for(c)) {
+    case 1: hipLaunchKernelGGL(bn_stats_v2_kernel<1>, g, b, 0, stream, x,
+                               acc, m, c, slot_mask); break;
+    case 2: hipLaunchKernelGGL(bn_stats_v2_kernel<2>, g, b, 0, stream, x,
+                               acc, m, c, slot_mask); break;
+    case 4: hipLaunchKernelGGL(bn_stats_v2_kernel<4>, g, b, 0, stream, x,
+                               acc, m, c, slot_mask); break;
+    default: hipLaunchKernelGGL(bn_stats_v2_kernel<8>, g, b, 0, stream, x,
+                                acc, m, c, slot_mask); break;
+  }
+}

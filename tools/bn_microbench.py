@@ -1,0 +1,84 @@
+#!/usr/bin/env python3
+"""BN stats-kernel A/B microbench (round-2 BN bandwidth track).
+
+Measures bn_stats (v1: 4-row unroll, serial tail, auto grid) against
+bn_stats_v2 (8-row unroll, shfl tail) over a grid sweep on the ResNet-50
+BYOL BN shapes at bs=512, and prints effective read bandwidth.  Run on the
+GPU box:
+
+    python tools/bn_microbench.py [--iters 20]
+"""
+
+import argparse
+import time
+
+import torch
+
+# (rows, channels) at bs=512 224px: conv BNs see B*H*W rows; head BN1d 2B
+SHAPES = [
+    (512 * 112 * 112, 64),   # stem
+    (512 * 56 * 56, 64),
+    (512 * 56 * 56, 256),
+    (512 * 28 * 28, 512),
+    (512 * 14 * 14, 1024),
+    (512 * 7 * 7, 2048),
+    (1024, 4096),            # projector/predictor BN1d
+]
+
+NSLOTS = 64
+
+
+def bench(fn, iters):
+    for _ in range(3):
+        fn()
+    torch.cuda.synchronize()
+    t = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t) / iters
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--iters", type=int, default=20)
+    args = ap.parse_args()
+    from byol_amd.ops import require_extension
+    ext = require_extension("bn microbench")
+
+    print(f"{'rows x C':>22} {'v1 ms':>8} {'v1 TB/s':>8} "
+          f"{'best v2 ms':>10} {'TB/s':>6} {'grid':>6}")
+    for m, c in SHAPES:
+        x = torch.randn(m, c, device="cuda")
+        acc = torch.zeros(NSLOTS * 2 * c, device="cuda")
+        bytes_read = m * c * 4
+
+        t1 = bench(lambda: ext.bn_stats(x, acc, m, c, NSLOTS), args.iters)
+
+        # correctness pin for v2 (any grid): sums must match v1's
+        ref = torch.zeros(2 * c, device="cuda")
+        acc.zero_()
+        ext.bn_stats(x, acc, m, c, NSLOTS)
+        ext.bn_reduce_slots(acc, ref, NSLOTS)
+
+        best = (None, None)
+        for grid in (1024, 2048, 4096, 8192):
+            acc.zero_()
+            ext.bn_stats_v2(x, acc, m, c, NSLOTS, grid)
+            got = torch.zeros(2 * c, device="cuda")
+            ext.bn_reduce_slots(acc, got, NSLOTS)
+            assert torch.allclose(got, ref, rtol=1e-4, atol=1e-2), \
+                (m, c, grid, (got - ref).abs().max().item())
+            t2 = bench(lambda g=grid: ext.bn_stats_v2(x, acc, m, c,
+                                                      NSLOTS, g),
+                       args.iters)
+            if best[0] is None or t2 < best[0]:
+                best = (t2, grid)
+        print(f"{m:>14} x {c:<5} {t1 * 1e3:>8.3f} "
+              f"{bytes_read / t1 / 1e12:>8.2f} {best[0] * 1e3:>10.3f} "
+              f"{bytes_read / best[0] / 1e12:>6.2f} {best[1]:>6}")
+        del x, acc
+
+
+if __name__ == "__main__":
+    main()
