@@ -278,6 +278,16 @@ void conv1x1_wgrad_kernel(const float* __restrict__ dY,
 #define LDSW_B 4096   // 32x128 floats
 #define LDSW_BUF (LDSW_A + LDSW_B)
 
+// XCD-aware bijective blockIdx remap (guide T1): consecutive logical tiles
+// share the A row-panel; grouping them per XCD makes those panel re-reads
+// L2 hits.  Bijective also when gridDim.x % 8 != 0.
+__device__ __forceinline__ int xcd_remap(int bid, int nwg) {
+  const int q = nwg >> 3, r = nwg & 7;
+  const int xcd = bid & 7, orig = bid >> 3;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig;
+}
+
+template <bool XSWZ, bool PRIO>
 __global__ __launch_bounds__(256)
 void conv1x1_gemm_fast_kernel(const float* __restrict__ A,
                               const float* __restrict__ B,
@@ -286,8 +296,9 @@ void conv1x1_gemm_fast_kernel(const float* __restrict__ A,
   __shared__ __attribute__((aligned(16))) float lds[2 * LDSW_BUF];
 
   const int ntiles_n = N / BN;
-  const int tile_m = blockIdx.x / ntiles_n;
-  const int tile_n = blockIdx.x % ntiles_n;
+  const int bid = XSWZ ? xcd_remap(blockIdx.x, gridDim.x) : blockIdx.x;
+  const int tile_m = bid / ntiles_n;
+  const int tile_n = bid % ntiles_n;
   const int64_t m0 = (int64_t)tile_m * BM;
   const int n0 = tile_n * BN;
 
@@ -344,6 +355,7 @@ void conv1x1_gemm_fast_kernel(const float* __restrict__ A,
     if (kt + 1 < ntiles_k) stage(buf ^ 1, (kt + 1) * BK);
     const float* la = &lds[buf * LDSW_BUF];
     const float* lb = &lds[buf * LDSW_BUF + LDSW_A];
+    if (PRIO) __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int kk = 0; kk < BK; kk += 2) {
       const int krow = kk + (lane >> 5);
@@ -361,6 +373,7 @@ void conv1x1_gemm_fast_kernel(const float* __restrict__ A,
         }
       }
     }
+    if (PRIO) __builtin_amdgcn_s_setprio(0);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     buf ^= 1;
@@ -394,6 +407,123 @@ void conv1x1_gemm_fast_kernel(const float* __restrict__ A,
     const float4 v = *reinterpret_cast<const float4*>(&cw[lrow * 64 + lcol]);
     *reinterpret_cast<float4*>(
         &Cmat[(gm_base + lrow) * N + gn_base + lcol]) = v;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 128x64 fast variant for N=64-class output widths (layer1 1x1s and the
+// dgrads whose output channel dim is 64/192): same glds 2-phase pipeline,
+// 4 waves each owning a 32x64 strip (acc 1x2 of 32x32).  LDS/buf =
+// 16 KB A + 8 KB B -> 3 workgroups/CU fit for extra block-level overlap.
+// ---------------------------------------------------------------------------
+#define BN64 64
+#define LDSW64_B 2048  // 32x64 floats
+#define LDSW64_BUF (LDSW_A + LDSW64_B)
+
+template <bool XSWZ, bool PRIO>
+__global__ __launch_bounds__(256)
+void conv1x1_gemm_fast64_kernel(const float* __restrict__ A,
+                                const float* __restrict__ B,
+                                float* __restrict__ Cmat,
+                                int64_t M, int Kd, int N) {
+  __shared__ __attribute__((aligned(16))) float lds[2 * LDSW64_BUF];
+
+  const int ntiles_n = N / BN64;
+  const int bid = XSWZ ? xcd_remap(blockIdx.x, gridDim.x) : blockIdx.x;
+  const int tile_m = bid / ntiles_n;
+  const int tile_n = bid % ntiles_n;
+  const int64_t m0 = (int64_t)tile_m * BM;
+  const int n0 = tile_n * BN64;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = wave * 32;  // each wave: rows [wr, wr+32) x all 64 cols
+
+  f32x16 acc[2];
+  #pragma unroll
+  for (int u = 0; u < 2; ++u)
+    #pragma unroll
+    for (int e = 0; e < 16; ++e) acc[u][e] = 0.f;
+
+  auto stage = [&](int buf, int k0) {
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {  // A: 16 KB, quad-rotated source swizzle
+      const int off = (wave * 4 + i) * 256 + lane * 4;
+      const int r = off >> 5;
+      const int q = (off & 31) >> 2;
+      const float* src =
+          A + (m0 + r) * (int64_t)Kd + k0 + (((q - r) & 7) << 2);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * LDSW64_BUF + (wave * 4 + i) * 256],
+          16, 0, 0);
+    }
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {  // B: 8 KB, [32][64] linear
+      const int off = (wave * 2 + i) * 256 + lane * 4;
+      const int r = off >> 6;
+      const int c = off & 63;
+      const float* src = B + (int64_t)(k0 + r) * N + n0 + c;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * LDSW64_BUF + LDSW_A + (wave * 2 + i) * 256],
+          16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  const int ntiles_k = Kd / BK;
+  int buf = 0;
+  for (int kt = 0; kt < ntiles_k; ++kt) {
+    if (kt + 1 < ntiles_k) stage(buf ^ 1, (kt + 1) * BK);
+    const float* la = &lds[buf * LDSW64_BUF];
+    const float* lb = &lds[buf * LDSW64_BUF + LDSW_A];
+    if (PRIO) __builtin_amdgcn_s_setprio(1);
+    #pragma unroll
+    for (int kk = 0; kk < BK; kk += 2) {
+      const int krow = kk + (lane >> 5);
+      const int row = wr + (lane & 31);
+      const float a = la[row * 32 + ((((krow >> 2) + row) & 7) << 2)
+                         + (krow & 3)];
+      #pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        const float b = lb[krow * BN64 + u * 32 + (lane & 31)];
+        acc[u] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc[u], 0, 0, 0);
+      }
+    }
+    if (PRIO) __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+
+  // epilogue: restage each wave's 32x64 strip through its LDS slice,
+  // store contiguous dwordx4 rows
+  __syncthreads();
+  float* cw = &lds[wave * 2048];
+  #pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    #pragma unroll
+    for (int e = 0; e < 16; ++e) {
+      const int lrow = (e & 3) + 8 * (e >> 2) + 4 * (lane >> 5);
+      const int lcol = u * 32 + (lane & 31);
+      cw[lrow * 64 + lcol] = acc[u][e];
+    }
+  }
+  __builtin_amdgcn_s_barrier();
+  const int64_t gm_base = m0 + wr;
+  #pragma unroll
+  for (int p = 0; p < 8; ++p) {
+    const int lrow = p * 4 + (lane >> 4);
+    const int lcol = (lane & 15) * 4;
+    const float4 v = *reinterpret_cast<const float4*>(&cw[lrow * 64 + lcol]);
+    *reinterpret_cast<float4*>(
+        &Cmat[(gm_base + lrow) * N + n0 + lcol]) = v;
   }
 }
 
@@ -550,14 +680,52 @@ __global__ void conv1x1_wgrad_reduce_kernel(const float* __restrict__ partial,
 
 static inline int64_t cdiv(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
+// A/B experiment knobs (read once per process; tools/conv_microbench.py
+// sweeps them): BYOL_CONV_XSWZ = XCD-aware tile remap (T1),
+// BYOL_CONV_PRIO = s_setprio around the MFMA cluster (T5).
+static bool conv_env_flag(const char* name, bool dflt) {
+  const char* v = getenv(name);
+  return v ? v[0] == '1' : dflt;
+}
+static bool conv_xswz() {
+  static bool f = conv_env_flag("BYOL_CONV_XSWZ", false);
+  return f;
+}
+static bool conv_prio() {
+  static bool f = conv_env_flag("BYOL_CONV_PRIO", false);
+  return f;
+}
+
+#define LAUNCH_FAST(KERN, GRID, ...)                                       \
+  do {                                                                     \
+    const bool xs = conv_xswz(), pr = conv_prio();                         \
+    if (xs && pr)                                                          \
+      hipLaunchKernelGGL((KERN<true, true>), GRID, dim3(256), 0, stream,   \
+                         __VA_ARGS__);                                     \
+    else if (xs)                                                           \
+      hipLaunchKernelGGL((KERN<true, false>), GRID, dim3(256), 0, stream,  \
+                         __VA_ARGS__);                                     \
+    else if (pr)                                                           \
+      hipLaunchKernelGGL((KERN<false, true>), GRID, dim3(256), 0, stream,  \
+                         __VA_ARGS__);                                     \
+    else                                                                   \
+      hipLaunchKernelGGL((KERN<false, false>), GRID, dim3(256), 0, stream, \
+                         __VA_ARGS__);                                     \
+  } while (0)
+
 void launch_conv1x1_fwd(const float* x, const float* w, const float* wt,
                         float* y, int64_t m, int k, int n,
                         hipStream_t stream) {
-  const int64_t grid = cdiv(m, BM) * cdiv(n, BN);
-  if (m % BM == 0 && n % BN == 0 && k % BK == 0 && wt != nullptr) {
-    hipLaunchKernelGGL(conv1x1_gemm_fast_kernel, dim3((uint32_t)grid),
-                       dim3(256), 0, stream, x, wt, y, m, k, n);
+  if (m % BM == 0 && k % BK == 0 && wt != nullptr && n % BN == 0) {
+    const int64_t grid = cdiv(m, BM) * cdiv(n, BN);
+    LAUNCH_FAST(conv1x1_gemm_fast_kernel, dim3((uint32_t)grid), x, wt, y, m,
+                k, n);
+  } else if (m % BM == 0 && k % BK == 0 && wt != nullptr && n % BN64 == 0) {
+    const int64_t grid = cdiv(m, BM) * cdiv(n, BN64);
+    LAUNCH_FAST(conv1x1_gemm_fast64_kernel, dim3((uint32_t)grid), x, wt, y,
+                m, k, n);
   } else {
+    const int64_t grid = cdiv(m, BM) * cdiv(n, BN);
     hipLaunchKernelGGL((conv1x1_gemm_kernel<true>), dim3((uint32_t)grid),
                        dim3(256), 0, stream, x, w, y, m, k, n);
   }
@@ -566,11 +734,16 @@ void launch_conv1x1_fwd(const float* x, const float* w, const float* wt,
 void launch_conv1x1_dgrad(const float* dy, const float* w, float* dx,
                           int64_t m, int n, int k, hipStream_t stream) {
   // dX[M,K] = dY[M,N] . W[N,K]  (A=dY, Kd=N, output N-dim = K)
-  const int64_t grid = cdiv(m, BM) * cdiv(k, BN);
-  if (m % BM == 0 && k % BN == 0 && n % BK == 0) {
-    hipLaunchKernelGGL(conv1x1_gemm_fast_kernel, dim3((uint32_t)grid),
-                       dim3(256), 0, stream, dy, w, dx, m, n, k);
+  if (m % BM == 0 && n % BK == 0 && k % BN == 0) {
+    const int64_t grid = cdiv(m, BM) * cdiv(k, BN);
+    LAUNCH_FAST(conv1x1_gemm_fast_kernel, dim3((uint32_t)grid), dy, w, dx,
+                m, n, k);
+  } else if (m % BM == 0 && n % BK == 0 && k % BN64 == 0) {
+    const int64_t grid = cdiv(m, BM) * cdiv(k, BN64);
+    LAUNCH_FAST(conv1x1_gemm_fast64_kernel, dim3((uint32_t)grid), dy, w, dx,
+                m, n, k);
   } else {
+    const int64_t grid = cdiv(m, BM) * cdiv(k, BN);
     hipLaunchKernelGGL((conv1x1_gemm_kernel<false>), dim3((uint32_t)grid),
                        dim3(256), 0, stream, dy, w, dx, m, n, k);
   }
