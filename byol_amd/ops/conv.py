@@ -175,7 +175,7 @@ class _Conv3x3Fn(torch.autograd.Function):
         y = torch.empty((b, n, ho, wo), device=x.device, dtype=x.dtype,
                         memory_format=torch.channels_last)
         m = b * ho * wo
-        if (m % 128 == 0 and n % 128 == 0 and k % 32 == 0
+        if (m % 128 == 0 and n % 64 == 0 and k % 32 == 0
                 and os.environ.get("BYOL_CONV3X3_FAST", "1") == "1"):
             # glds-pipelined fast path over a zero-padded copy (guard-free
             # taps); the pad costs one extra read+write of x

@@ -41,3 +41,49 @@ static inline int grid_1d(int64_t n, int block, int max_blocks = 16384) {
   if (g < 1) g = 1;
   return static_cast<int>(g);
 }
+
+// ---------------------------------------------------------------------------
+// shared conv-kernel experiment plumbing (conv1x1.hip / conv3x3.hip)
+// ---------------------------------------------------------------------------
+
+// XCD-aware bijective blockIdx remap (guide T1): consecutive logical tiles
+// share an operand panel; grouping them per XCD makes panel re-reads L2
+// hits.  Bijective also when nwg % 8 != 0.
+__device__ __forceinline__ int xcd_remap(int bid, int nwg) {
+  const int q = nwg >> 3, r = nwg & 7;
+  const int xcd = bid & 7, orig = bid >> 3;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig;
+}
+
+// A/B knobs, read once per process (tools/conv_microbench.py sweeps them
+// across processes): BYOL_CONV_XSWZ = XCD tile remap (T1), BYOL_CONV_PRIO =
+// s_setprio around the MFMA cluster (T5).
+static inline bool conv_env_flag(const char* name, bool dflt) {
+  const char* v = getenv(name);
+  return v ? v[0] == '1' : dflt;
+}
+static inline bool conv_xswz() {
+  static bool f = conv_env_flag("BYOL_CONV_XSWZ", false);
+  return f;
+}
+static inline bool conv_prio() {
+  static bool f = conv_env_flag("BYOL_CONV_PRIO", false);
+  return f;
+}
+
+#define LAUNCH_FAST(KERN, GRID, ...)                                       \
+  do {                                                                     \
+    const bool xs = conv_xswz(), pr = conv_prio();                         \
+    if (xs && pr)                                                          \
+      hipLaunchKernelGGL((KERN<true, true>), GRID, dim3(256), 0, stream,   \
+                         __VA_ARGS__);                                     \
+    else if (xs)                                                           \
+      hipLaunchKernelGGL((KERN<true, false>), GRID, dim3(256), 0, stream,  \
+                         __VA_ARGS__);                                     \
+    else if (pr)                                                           \
+      hipLaunchKernelGGL((KERN<false, true>), GRID, dim3(256), 0, stream,  \
+                         __VA_ARGS__);                                     \
+    else                                                                   \
+      hipLaunchKernelGGL((KERN<false, false>), GRID, dim3(256), 0, stream, \
+                         __VA_ARGS__);                                     \
+  } while (0)

@@ -278,15 +278,6 @@ void conv1x1_wgrad_kernel(const float* __restrict__ dY,
 #define LDSW_B 4096   // 32x128 floats
 #define LDSW_BUF (LDSW_A + LDSW_B)
 
-// XCD-aware bijective blockIdx remap (guide T1): consecutive logical tiles
-// share the A row-panel; grouping them per XCD makes those panel re-reads
-// L2 hits.  Bijective also when gridDim.x % 8 != 0.
-__device__ __forceinline__ int xcd_remap(int bid, int nwg) {
-  const int q = nwg >> 3, r = nwg & 7;
-  const int xcd = bid & 7, orig = bid >> 3;
-  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig;
-}
-
 template <bool XSWZ, bool PRIO>
 __global__ __launch_bounds__(256)
 void conv1x1_gemm_fast_kernel(const float* __restrict__ A,
@@ -679,39 +670,6 @@ __global__ void conv1x1_wgrad_reduce_kernel(const float* __restrict__ partial,
 }
 
 static inline int64_t cdiv(int64_t a, int64_t b) { return (a + b - 1) / b; }
-
-// A/B experiment knobs (read once per process; tools/conv_microbench.py
-// sweeps them): BYOL_CONV_XSWZ = XCD-aware tile remap (T1),
-// BYOL_CONV_PRIO = s_setprio around the MFMA cluster (T5).
-static bool conv_env_flag(const char* name, bool dflt) {
-  const char* v = getenv(name);
-  return v ? v[0] == '1' : dflt;
-}
-static bool conv_xswz() {
-  static bool f = conv_env_flag("BYOL_CONV_XSWZ", false);
-  return f;
-}
-static bool conv_prio() {
-  static bool f = conv_env_flag("BYOL_CONV_PRIO", false);
-  return f;
-}
-
-#define LAUNCH_FAST(KERN, GRID, ...)                                       \
-  do {                                                                     \
-    const bool xs = conv_xswz(), pr = conv_prio();                         \
-    if (xs && pr)                                                          \
-      hipLaunchKernelGGL((KERN<true, true>), GRID, dim3(256), 0, stream,   \
-                         __VA_ARGS__);                                     \
-    else if (xs)                                                           \
-      hipLaunchKernelGGL((KERN<true, false>), GRID, dim3(256), 0, stream,  \
-                         __VA_ARGS__);                                     \
-    else if (pr)                                                           \
-      hipLaunchKernelGGL((KERN<false, true>), GRID, dim3(256), 0, stream,  \
-                         __VA_ARGS__);                                     \
-    else                                                                   \
-      hipLaunchKernelGGL((KERN<false, false>), GRID, dim3(256), 0, stream, \
-                         __VA_ARGS__);                                     \
-  } while (0)
 
 void launch_conv1x1_fwd(const float* x, const float* w, const float* wt,
                         float* y, int64_t m, int k, int n,

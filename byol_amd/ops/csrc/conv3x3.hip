@@ -177,6 +177,7 @@ __global__ void pad_nhwc_kernel(const float* __restrict__ x,
 #define C3F_LDSW_B 4096
 #define C3F_LDSW_BUF (C3F_LDSW_A + C3F_LDSW_B)
 
+template <bool XSWZ, bool PRIO>
 __global__ __launch_bounds__(256)
 void conv3x3_fwd_fast_kernel(const float* __restrict__ Xp,  // padded NHWC
                              const float* __restrict__ Wp,  // [9][K][N]
@@ -187,8 +188,9 @@ void conv3x3_fwd_fast_kernel(const float* __restrict__ Xp,  // padded NHWC
 
   const int64_t M = (int64_t)B * Ho * Wo;
   const int ntiles_n = N / C3_BN;
-  const int tile_m = blockIdx.x / ntiles_n;
-  const int tile_n = blockIdx.x % ntiles_n;
+  const int bid = XSWZ ? xcd_remap(blockIdx.x, gridDim.x) : blockIdx.x;
+  const int tile_m = bid / ntiles_n;
+  const int tile_n = bid % ntiles_n;
   const int64_t m0 = (int64_t)tile_m * C3_BM;
   const int n0 = tile_n * C3_BN;
 
@@ -266,6 +268,7 @@ void conv3x3_fwd_fast_kernel(const float* __restrict__ Xp,  // padded NHWC
     if (step + 1 < nsteps) stage(buf ^ 1, step + 1);
     const float* la = &lds[buf * C3F_LDSW_BUF];
     const float* lb = &lds[buf * C3F_LDSW_BUF + C3F_LDSW_A];
+    if (PRIO) __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int kk = 0; kk < C3_BK; kk += 2) {
       const int krow = kk + (lane >> 5);
@@ -282,6 +285,7 @@ void conv3x3_fwd_fast_kernel(const float* __restrict__ Xp,  // padded NHWC
         }
       }
     }
+    if (PRIO) __builtin_amdgcn_s_setprio(0);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     buf ^= 1;
@@ -313,6 +317,138 @@ void conv3x3_fwd_fast_kernel(const float* __restrict__ Xp,  // padded NHWC
   }
 }
 
+// 128x64 fast variant for N=64-class widths (layer1's 3x3 at C=64 — the
+// shape where the guarded slow kernel was 2x MIOpen).  Same padded-glds
+// pipeline; 4 waves each own a 32x64 strip; LDS/buf = 16 KB A + 8 KB B.
+#define C3F64_LDSW_B 2048
+#define C3F64_LDSW_BUF (C3F_LDSW_A + C3F64_LDSW_B)
+
+template <bool XSWZ, bool PRIO>
+__global__ __launch_bounds__(256)
+void conv3x3_fwd_fast64_kernel(const float* __restrict__ Xp,
+                               const float* __restrict__ Wp,  // [9][K][N]
+                               float* __restrict__ Y,
+                               int B, int Hp, int Wpp,
+                               int Ho, int Wo, int K, int N, int stride) {
+  __shared__ __attribute__((aligned(16))) float lds[2 * C3F64_LDSW_BUF];
+
+  const int64_t M = (int64_t)B * Ho * Wo;
+  const int ntiles_n = N / 64;
+  const int bid = XSWZ ? xcd_remap(blockIdx.x, gridDim.x) : blockIdx.x;
+  const int tile_m = bid / ntiles_n;
+  const int tile_n = bid % ntiles_n;
+  const int64_t m0 = (int64_t)tile_m * C3_BM;
+  const int n0 = tile_n * 64;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = wave * 32;
+
+  const float* abase[4];
+  int aq[4];
+  #pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int off = (wave * 4 + i) * 256 + lane * 4;
+    const int r = off >> 5;
+    const int q = (off & 31) >> 2;
+    aq[i] = ((q - r) & 7) << 2;
+    const int64_t gm = m0 + r;
+    const int64_t howo = (int64_t)Ho * Wo;
+    const int bb = (int)(gm / howo);
+    const int rem = (int)(gm % howo);
+    const int iy = (rem / Wo) * stride;
+    const int ix = (rem % Wo) * stride;
+    abase[i] = Xp + (((int64_t)bb * Hp + iy) * Wpp + ix) * K;
+  }
+
+  f32x16_c3 acc[2];
+  #pragma unroll
+  for (int u = 0; u < 2; ++u)
+    #pragma unroll
+    for (int e = 0; e < 16; ++e) acc[u][e] = 0.f;
+
+  const int nk = K / C3_BK;
+  const int nsteps = 9 * nk;
+
+  auto stage = [&](int buf, int step) {
+    const int tap = step / nk;
+    const int k0 = (step % nk) * C3_BK;
+    const int dy = tap / 3, dx = tap % 3;
+    const int64_t tap_off = ((int64_t)dy * Wpp + dx) * K;
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const float* src = abase[i] + tap_off + k0 + aq[i];
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * C3F64_LDSW_BUF + (wave * 4 + i) * 256],
+          16, 0, 0);
+    }
+    const float* wtap = Wp + ((int64_t)tap * K + k0) * N;
+    #pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int off = (wave * 2 + i) * 256 + lane * 4;
+      const int r = off >> 6;
+      const int c = off & 63;
+      const float* src = wtap + (int64_t)r * N + n0 + c;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * C3F64_LDSW_BUF + C3F_LDSW_A + (wave * 2 + i) * 256],
+          16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int buf = 0;
+  for (int step = 0; step < nsteps; ++step) {
+    if (step + 1 < nsteps) stage(buf ^ 1, step + 1);
+    const float* la = &lds[buf * C3F64_LDSW_BUF];
+    const float* lb = &lds[buf * C3F64_LDSW_BUF + C3F_LDSW_A];
+    if (PRIO) __builtin_amdgcn_s_setprio(1);
+    #pragma unroll
+    for (int kk = 0; kk < C3_BK; kk += 2) {
+      const int krow = kk + (lane >> 5);
+      const int row = wr + (lane & 31);
+      const float a = la[row * 32 + ((((krow >> 2) + row) & 7) << 2)
+                         + (krow & 3)];
+      #pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        const float b = lb[krow * 64 + u * 32 + (lane & 31)];
+        acc[u] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc[u], 0, 0, 0);
+      }
+    }
+    if (PRIO) __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+
+  __syncthreads();
+  float* cw = &lds[wave * 2048];
+  #pragma unroll
+  for (int u = 0; u < 2; ++u)
+    #pragma unroll
+    for (int e = 0; e < 16; ++e) {
+      const int lrow = (e & 3) + 8 * (e >> 2) + 4 * (lane >> 5);
+      const int lcol = u * 32 + (lane & 31);
+      cw[lrow * 64 + lcol] = acc[u][e];
+    }
+  __builtin_amdgcn_s_barrier();
+  const int64_t gm_base = m0 + wr;
+  #pragma unroll
+  for (int p = 0; p < 8; ++p) {
+    const int lrow = p * 4 + (lane >> 4);
+    const int lcol = (lane & 15) * 4;
+    const float4 v = *reinterpret_cast<const float4*>(&cw[lrow * 64 + lcol]);
+    *reinterpret_cast<float4*>(
+        &Y[(gm_base + lrow) * N + n0 + lcol]) = v;
+  }
+}
+
 void launch_pad_nhwc(const float* x, float* xp, int b, int hi, int wi,
                      int c, hipStream_t stream) {
   const int64_t total = (int64_t)b * (hi + 2) * (wi + 2) * (c >> 2);
@@ -324,10 +460,15 @@ void launch_conv3x3_fwd_fast(const float* xp, const float* wp, float* y,
                              int b, int hi, int wi, int ho, int wo, int k,
                              int n, int stride, hipStream_t stream) {
   const int64_t m = (int64_t)b * ho * wo;
-  const int64_t grid = (m / C3_BM) * (n / C3_BN);
-  hipLaunchKernelGGL(conv3x3_fwd_fast_kernel, dim3((uint32_t)grid),
-                     dim3(256), 0, stream, xp, wp, y, b, hi + 2, wi + 2, ho,
-                     wo, k, n, stride);
+  if (n % C3_BN == 0) {
+    const int64_t grid = (m / C3_BM) * (n / C3_BN);
+    LAUNCH_FAST(conv3x3_fwd_fast_kernel, dim3((uint32_t)grid), xp, wp, y, b,
+                hi + 2, wi + 2, ho, wo, k, n, stride);
+  } else {
+    const int64_t grid = (m / C3_BM) * (n / 64);
+    LAUNCH_FAST(conv3x3_fwd_fast64_kernel, dim3((uint32_t)grid), xp, wp, y,
+                b, hi + 2, wi + 2, ho, wo, k, n, stride);
+  }
 }
 
 void launch_conv3x3_fwd(const float* x, const float* wp, float* y, int b,
