@@ -132,6 +132,11 @@ def build_parser() -> argparse.ArgumentParser:
                         help="NHWC memory format for conv throughput")
     parser.add_argument("--perf-stats", action="store_true", default=False,
                         help="per-phase HIP-event timing summary per epoch")
+    parser.add_argument("--hip-graph", action="store_true", default=False,
+                        help="capture the training step as one hipGraph and "
+                             "replay it (single-GPU, no --polyak-ema; "
+                             "removes ~2k kernel-launch boundaries per "
+                             "step of host gap time)")
     return parser
 
 

@@ -202,6 +202,46 @@ def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
     timer = PhaseTimer(enabled=getattr(args, "perf_stats", False),
                        use_cuda=args.cuda)
 
+    from ..ops.classifier import cross_entropy_topk
+
+    def train_step_body(a1, a2, lab):
+        """One full training step (fwd + losses + bwd + opt) — runs eager
+        or captured into a hipGraph (``--hip-graph``); returns the five
+        scalar tensors the epoch aggregation reads."""
+        with autocast_ctx:
+            out = model(a1, a2)
+            byol_loss = loss_function(
+                online_prediction1=out["online_prediction1"].float(),
+                online_prediction2=out["online_prediction2"].float(),
+                target_projection1=out["target_projection1"].float(),
+                target_projection2=out["target_projection2"].float())
+            # fused CE + top-1/top-5 in one HIP kernel on GPU
+            # (byol_amd/ops/classifier.py); composed oracle elsewhere
+            classifier_loss, acc1, acc5 = cross_entropy_topk(
+                out["linear_preds"].float().contiguous(),
+                torch.cat([lab, lab], 0))
+            total = byol_loss + classifier_loss
+        optimizer.zero_grad()
+        total.backward()
+        if is_ddp:
+            model.finish_grad_sync()
+        if args.clip > 0:
+            nn.utils.clip_grad_value_(model.parameters(), args.clip)
+        optimizer.step()
+        if args.polyak_ema > 0:
+            layers.polyak_ema_parameters(model, args.polyak_ema)
+        return total, byol_loss, classifier_loss, acc1, acc5
+
+    # hipGraph path: single-GPU training without polyak (the graph wrapper
+    # owns the EMA decay/LR device scalars; DDP+graph is a later
+    # experiment).  First GRAPH_WARM minibatches run eager (MIOpen finds,
+    # fused-LARS state), then the step is captured once and replayed.
+    graph_wanted = (not is_eval and args.cuda
+                    and getattr(args, "hip_graph", False)
+                    and args.polyak_ema == 0 and not is_ddp)
+    GRAPH_WARM = 2
+    graphed = None
+
     timer.start("data")
     for num_minibatches, (augmentation1, augmentation2, labels) in \
             enumerate(loader):
@@ -211,61 +251,75 @@ def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
             labels = labels.cuda(non_blocking=True)
         timer.stop()
 
-        timer.start("forward")
-        with torch.no_grad() if is_eval else _dummy_context():
-            with autocast_ctx:
-                if is_eval and args.polyak_ema > 0:
-                    output_dict = layers.get_polyak_prediction(
-                        model, pred_fn=functools.partial(
-                            model, augmentation1, augmentation2))
-                else:
-                    output_dict = model(augmentation1, augmentation2)
+        if not is_eval:
+            timer.start("step")
+            stepped = False
+            if graph_wanted and num_minibatches >= GRAPH_WARM:
+                if graphed is None:
+                    from .graph_step import GraphedTrainStep
+                    static = (augmentation1.clone(), augmentation2.clone(),
+                              labels.clone())
+                    cand = GraphedTrainStep(model, optimizer,
+                                            train_step_body, static,
+                                            warmup_steps=0)
+                    try:
+                        cand.capture()
+                        graphed = cand
+                    except Exception as exc:  # noqa: BLE001 — eager works
+                        graph_wanted = False
+                        print(f"hip-graph capture failed ({exc}); "
+                              "continuing eager")
+                if graphed is not None and \
+                        augmentation1.shape == graphed.static_inputs[0].shape:
+                    vals = graphed.replay(augmentation1, augmentation2,
+                                          labels)
+                    stepped = True
+            if not stepped:
+                vals = train_step_body(augmentation1, augmentation2, labels)
+            total, byol_loss, classifier_loss, acc1, acc5 = vals
+            timer.stop()
+        else:
+            timer.start("forward")
+            with torch.no_grad():
+                with autocast_ctx:
+                    if args.polyak_ema > 0:
+                        output_dict = layers.get_polyak_prediction(
+                            model, pred_fn=functools.partial(
+                                model, augmentation1, augmentation2))
+                    else:
+                        output_dict = model(augmentation1, augmentation2)
+                    byol_loss = loss_function(
+                        online_prediction1=output_dict[
+                            "online_prediction1"].float(),
+                        online_prediction2=output_dict[
+                            "online_prediction2"].float(),
+                        target_projection1=output_dict[
+                            "target_projection1"].float(),
+                        target_projection2=output_dict[
+                            "target_projection2"].float())
+                    classifier_loss, acc1, acc5 = cross_entropy_topk(
+                        output_dict["linear_preds"].float().contiguous(),
+                        labels)
+                    total = byol_loss + classifier_loss
+            timer.stop()
 
-                timer.stop()
-                timer.start("loss")
-                byol_loss = loss_function(
-                    online_prediction1=output_dict["online_prediction1"].float(),
-                    online_prediction2=output_dict["online_prediction2"].float(),
-                    target_projection1=output_dict["target_projection1"].float(),
-                    target_projection2=output_dict["target_projection2"].float())
-                classifier_labels = labels if is_eval else \
-                    torch.cat([labels, labels], 0)
-                # fused CE + top-1/top-5 in one HIP kernel on GPU
-                # (byol_amd/ops/classifier.py); composed oracle elsewhere
-                from ..ops.classifier import cross_entropy_topk
-                classifier_loss, acc1, acc5 = cross_entropy_topk(
-                    output_dict["linear_preds"].float().contiguous(),
-                    classifier_labels)
+        with torch.no_grad():
             loss_t = {
-                "loss_mean": byol_loss + classifier_loss,
+                "loss_mean": total,
                 "byol_loss_mean": byol_loss,
                 "linear_loss_mean": classifier_loss,
                 "top1_mean": acc1,
                 "top5_mean": acc5,
             }
-            detached_t = {k: (v.detach() if torch.is_tensor(v) else v)
+            # clone: under --hip-graph these are the graph's static output
+            # tensors, whose storage is rewritten by the next replay
+            detached_t = {k: (v.detach().clone() if torch.is_tensor(v)
+                              else v)
                           for k, v in loss_t.items()}
             loss_map = detached_t if not loss_map else \
                 _sum_scalars(loss_map, detached_t)
             num_samples += augmentation1.size(0)
             timer.add_samples(augmentation1.size(0))
-        timer.stop()
-
-        if not is_eval:
-            timer.start("backward")
-            optimizer.zero_grad()
-            loss_t["loss_mean"].backward()
-            if is_ddp:
-                model.finish_grad_sync()
-            timer.stop()
-            timer.start("optimizer")
-            if args.clip > 0:
-                nn.utils.clip_grad_value_(model.parameters(), args.clip)
-            optimizer.step()
-            if args.polyak_ema > 0:
-                layers.polyak_ema_parameters(model, args.polyak_ema)
-            timer.stop()
-            del loss_t
 
         if args.debug_step:
             break
