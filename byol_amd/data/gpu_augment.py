@@ -178,13 +178,21 @@ class GPUTwoViewAugment:
         ext.aug_sample(src_nhwc.reshape(-1), dst.reshape(-1), gray_sum,
                        crop.reshape(-1), hs, ws, s, use_v2)
         ext.aug_color(dst.reshape(-1), gray_sum, cparam.reshape(-1), s)
+        if not self.dali_mode and sigma_np.any():
+            # per-sample-sigma separable blur: 3 HIP launches directly on
+            # the NHWC buffer (weights, vertical, horizontal) — replaces
+            # the composed B*C-group conv pair
+            sigma = torch.from_numpy(sigma_np).to(device, non_blocking=True)
+            tmp = torch.empty_like(dst)
+            blurred = torch.empty_like(dst)
+            wts = torch.empty(b, self.blur_kernel, device=device,
+                              dtype=torch.float32)
+            ext.aug_blur(dst.reshape(-1), tmp.reshape(-1),
+                         blurred.reshape(-1), sigma, wts.reshape(-1), s,
+                         self.blur_kernel)
+            dst = blurred
         # NHWC [B,S,S,3] -> channels_last NCHW view
         out = dst.permute(0, 3, 1, 2)
-        if not self.dali_mode and sigma_np.any():
-            sigma = torch.from_numpy(sigma_np).to(device, non_blocking=True)
-            out = _gaussian_blur_batched(
-                out.contiguous(memory_format=torch.contiguous_format),
-                sigma, self.blur_kernel)
         return out.contiguous(memory_format=torch.channels_last)
 
     def __call__(self, batch_nchw: torch.Tensor):

@@ -235,3 +235,92 @@ void launch_aug_color(float* img, const float* gray_sum, const float* cparam,
   hipLaunchKernelGGL(aug_color_kernel, dim3(grid_1d(total, 256)), dim3(256),
                      0, stream, img, gray_sum, cparam, b, s);
 }
+
+// ---------------------------------------------------------------------------
+// Per-sample-sigma separable Gaussian blur, NHWC [B,S,S,3] in [0,1].
+// Replaces the composed grouped-conv pair (B*C groups) the Python path used:
+// 3 launches total — weights, vertical pass, horizontal pass.  sigma[b]==0
+// rows get the identity kernel (pass-through).  Reflect padding, matching
+// torchvision GaussianBlur / data/transforms.py.
+// ---------------------------------------------------------------------------
+
+__global__ void blur_weights_kernel(const float* __restrict__ sigma,
+                                    float* __restrict__ wts,  // [B][ksize]
+                                    int b, int ksize) {
+  const int bi = blockIdx.x * blockDim.x + threadIdx.x;
+  if (bi >= b) return;
+  const float sg = sigma[bi];
+  float* w = wts + (int64_t)bi * ksize;
+  const int half = ksize / 2;
+  if (sg <= 0.f) {
+    for (int k = 0; k < ksize; ++k) w[k] = (k == half) ? 1.f : 0.f;
+    return;
+  }
+  float tot = 0.f;
+  for (int k = 0; k < ksize; ++k) {
+    const float d = (float)(k - half);
+    const float v = __expf(-d * d / (2.f * sg * sg));
+    w[k] = v;
+    tot += v;
+  }
+  const float inv = 1.f / tot;
+  for (int k = 0; k < ksize; ++k) w[k] *= inv;
+}
+
+// reflect index into [0, n): torchvision "reflect" (no edge repeat)
+__device__ __forceinline__ int reflect_idx(int i, int n) {
+  if (i < 0) i = -i;
+  if (i >= n) i = 2 * n - 2 - i;
+  return i;
+}
+
+// VERT = blur along y (else along x).  One thread per output element; the
+// 64-lane wave walks x*c contiguously so every tap row is a coalesced read.
+template <bool VERT>
+__global__ void blur_pass_kernel(const float* __restrict__ src,
+                                 float* __restrict__ dst,
+                                 const float* __restrict__ wts,
+                                 int b, int s, int ksize) {
+  const int64_t per_img = (int64_t)s * s * 3;
+  const int64_t total = (int64_t)b * per_img;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int half = ksize / 2;
+  const int rowlen = s * 3;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += stride) {
+    const int bi = (int)(i / per_img);
+    const int64_t rem = i % per_img;
+    const int y = (int)(rem / rowlen);
+    const int xc = (int)(rem % rowlen);
+    const float* w = wts + (int64_t)bi * ksize;
+    const float* img = src + (int64_t)bi * per_img;
+    float acc = 0.f;
+    if (VERT) {
+      #pragma unroll 4
+      for (int k = 0; k < ksize; ++k) {
+        const int yy = reflect_idx(y + k - half, s);
+        acc = fmaf(w[k], img[(int64_t)yy * rowlen + xc], acc);
+      }
+    } else {
+      const int x = xc / 3, c = xc - 3 * x;
+      #pragma unroll 4
+      for (int k = 0; k < ksize; ++k) {
+        const int xx = reflect_idx(x + k - half, s);
+        acc = fmaf(w[k], img[(int64_t)y * rowlen + xx * 3 + c], acc);
+      }
+    }
+    dst[i] = acc < 0.f ? 0.f : (acc > 1.f ? 1.f : acc);
+  }
+}
+
+void launch_aug_blur(const float* img, float* tmp, float* out,
+                     const float* sigma, float* wts, int b, int s,
+                     int ksize, hipStream_t stream) {
+  hipLaunchKernelGGL(blur_weights_kernel, dim3((b + 255) / 256), dim3(256),
+                     0, stream, sigma, wts, b, ksize);
+  const int64_t total = (int64_t)b * s * s * 3;
+  hipLaunchKernelGGL(blur_pass_kernel<true>, dim3(grid_1d(total, 256)),
+                     dim3(256), 0, stream, img, tmp, wts, b, s, ksize);
+  hipLaunchKernelGGL(blur_pass_kernel<false>, dim3(grid_1d(total, 256)),
+                     dim3(256), 0, stream, tmp, out, wts, b, s, ksize);
+}

@@ -63,6 +63,9 @@ void launch_aug_sample(const float* src, float* dst, float* gray_sum,
                        int use_v2, hipStream_t stream);
 void launch_aug_color(float* img, const float* gray_sum, const float* cparam,
                       int b, int s, hipStream_t stream);
+void launch_aug_blur(const float* img, float* tmp, float* out,
+                     const float* sigma, float* wts, int b, int s,
+                     int ksize, hipStream_t stream);
 void launch_ce_topk_fwd(const float* logits, const int64_t* labels,
                         float* out, float* row_stats, int m, int n,
                         hipStream_t stream);
@@ -302,6 +305,20 @@ void aug_color(torch::Tensor img, torch::Tensor gray_sum,
                    cparam.data_ptr<float>(), b, (int)s, stream);
 }
 
+void aug_blur(torch::Tensor img, torch::Tensor tmp, torch::Tensor out,
+              torch::Tensor sigma, torch::Tensor wts, int64_t s,
+              int64_t ksize) {
+  CHECK_IN(img); CHECK_IN(tmp); CHECK_IN(out); CHECK_IN(sigma);
+  CHECK_IN(wts);
+  const int b = (int)sigma.numel();
+  TORCH_CHECK(img.numel() == (int64_t)b * s * s * 3, "img shape mismatch");
+  TORCH_CHECK(wts.numel() >= (int64_t)b * ksize, "wts too small");
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_aug_blur(img.data_ptr<float>(), tmp.data_ptr<float>(),
+                  out.data_ptr<float>(), sigma.data_ptr<float>(),
+                  wts.data_ptr<float>(), b, (int)s, (int)ksize, stream);
+}
+
 void ce_topk_fwd(torch::Tensor logits, torch::Tensor labels,
                  torch::Tensor out, torch::Tensor row_stats) {
   CHECK_IN(logits); CHECK_IN(out); CHECK_IN(row_stats);
@@ -480,6 +497,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ce_bwd", &ce_bwd);
   mod.def("aug_sample", &aug_sample);
   mod.def("aug_color", &aug_color);
+  mod.def("aug_blur", &aug_blur);
   mod.def("bn_stats", &bn_stats);
   mod.def("bn_stats_v2", &bn_stats_v2);
   mod.def("bn_reduce_slots", &bn_reduce_slots);
