@@ -174,7 +174,9 @@ class GPUTwoViewAugment:
         cparam = torch.from_numpy(cparam_np).to(device, non_blocking=True)
         dst = torch.empty(b, s, s, 3, device=device, dtype=torch.float32)
         gray_sum = torch.zeros(b, device=device, dtype=torch.float32)
-        use_v2 = 1 if os.environ.get("BYOL_AUG_V2", "0") == "1" else 0
+        # v2 (sample-major blocks, one gray atomic per block) measured 7.9x
+        # v1 (79 -> 10 ms per 2-view bs=512 call, r2 call 2) — default ON
+        use_v2 = 1 if os.environ.get("BYOL_AUG_V2", "1") == "1" else 0
         ext.aug_sample(src_nhwc.reshape(-1), dst.reshape(-1), gray_sum,
                        crop.reshape(-1), hs, ws, s, use_v2)
         ext.aug_color(dst.reshape(-1), gray_sum, cparam.reshape(-1), s)
