@@ -85,6 +85,25 @@ def main():
         t_raw = timeit(lambda: C.conv1x1_fwd(xr, wv, wt, yr, m, cin, cout),
                        args.iters)
 
+        # raw backward split: our dgrad / wgrad kernels vs MIOpen's split
+        # convolution_backward (where exactly does backward lose?)
+        dyr = g.permute(0, 2, 3, 1).reshape(-1, cout).contiguous()
+        dxr = torch.empty(m, cin, device="cuda")
+        t_dg = timeit(lambda: C.conv1x1_dgrad(dyr, wv, dxr, m, cout, cin),
+                      args.iters)
+        dwr = torch.zeros(cout, cin, device="cuda")
+        def wg():
+            dwr.zero_()
+            C.conv1x1_wgrad(dyr, xr, dwr, m, cout, cin)
+        t_wg = timeit(wg, args.iters)
+        w4 = w.reshape(cout, cin, 1, 1)
+        t_dg_m = timeit(lambda: torch.ops.aten.convolution_backward(
+            g, x, w4, [0], [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+            [True, False, False]), args.iters)
+        t_wg_m = timeit(lambda: torch.ops.aten.convolution_backward(
+            g, x, w4, [0], [1, 1], [0, 0], [1, 1], False, [0, 0], 1,
+            [False, True, False]), args.iters)
+
         # perf: full fwd+bwd
         def bwd_miopen():
             xr = x.detach().requires_grad_(True)
@@ -101,7 +120,8 @@ def main():
         tag = "<<" if to_f < tm_f and to_b < tm_b else ""
         print(f"H{hgt:>3} K{cin:>5} N{cout:>5} "
               f"{tm_f:9.3f} {to_f:9.3f} raw{t_raw:8.3f} {tm_b:9.3f} "
-              f"{to_b:9.3f}  {ef:.2e} {edx:.2e} {edw:.2e} {tag}")
+              f"{to_b:9.3f}  {ef:.2e} {edx:.2e} {edw:.2e} {tag} "
+              f"dg{t_dg:7.3f}/{t_dg_m:7.3f} wg{t_wg:7.3f}/{t_wg_m:7.3f}")
 
 
 def main_3x3(batch=512, iters=8):
