@@ -20,6 +20,10 @@ void launch_bn_stats(const float* x, float* acc, int64_t m, int c,
                      int slot_mask, hipStream_t stream);
 void launch_bn_stats_v2(const float* x, float* acc, int64_t m, int c,
                         int slot_mask, int grid, hipStream_t stream);
+void launch_bn_bwd_reduce_v2(const float* dy, const float* y, const float* x,
+                             const float* mean, const float* invstd,
+                             float* red, int64_t m, int c, int relu,
+                             int slot_mask, int grid, hipStream_t stream);
 void launch_bn_reduce_slots(const float* in, float* out, int n2c, int nslots,
                             hipStream_t stream);
 void launch_bn_finalize(const float* acc, float* mean, float* invstd,
@@ -220,6 +224,19 @@ void bn_stats_v2(torch::Tensor x, torch::Tensor acc, int64_t m, int64_t c,
   auto stream = at::hip::getCurrentHIPStream();
   launch_bn_stats_v2(x.data_ptr<float>(), acc.data_ptr<float>(), m, (int)c,
                      (int)(nslots - 1), (int)grid, stream);
+}
+
+void bn_bwd_reduce_v2(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+                      torch::Tensor mean, torch::Tensor invstd,
+                      torch::Tensor red, int64_t m, int64_t c,
+                      int64_t relu, int64_t nslots, int64_t grid) {
+  CHECK_IN(dy); CHECK_IN(y); CHECK_IN(x); CHECK_IN(red);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_bwd_reduce_v2(dy.data_ptr<float>(), y.data_ptr<float>(),
+                          x.data_ptr<float>(), mean.data_ptr<float>(),
+                          invstd.data_ptr<float>(), red.data_ptr<float>(),
+                          m, (int)c, (int)relu, (int)(nslots - 1),
+                          (int)grid, stream);
 }
 
 void bn_reduce_slots(torch::Tensor in, torch::Tensor out, int64_t nslots) {
@@ -500,6 +517,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("aug_blur", &aug_blur);
   mod.def("bn_stats", &bn_stats);
   mod.def("bn_stats_v2", &bn_stats_v2);
+  mod.def("bn_bwd_reduce_v2", &bn_bwd_reduce_v2);
   mod.def("bn_reduce_slots", &bn_reduce_slots);
   mod.def("bn_finalize", &bn_finalize);
   mod.def("bn_apply", &bn_apply);

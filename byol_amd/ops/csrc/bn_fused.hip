@@ -567,3 +567,135 @@ void launch_bn_stats_v2(const float* x, float* acc, int64_t m, int c,
                                 acc, m, c, slot_mask); break;
   }
 }
+
+// bwd-reduce v2: 4-row unroll (12 independent loads in flight across the
+// dy/y/x streams) + the same shfl tail as stats v2.
+template <int KQ>
+__global__ void bn_bwd_reduce_v2_kernel(const float* __restrict__ dy,
+                                        const float* __restrict__ y,
+                                        const float* __restrict__ x,
+                                        const float* __restrict__ mean,
+                                        const float* __restrict__ invstd,
+                                        float* __restrict__ red,
+                                        int64_t m, int c, int relu,
+                                        int slot_mask) {
+  float* const red_slot = red + (int64_t)(blockIdx.x & slot_mask) * 2 * c;
+  const int c4 = c >> 2;
+  const int nthread = blockDim.x;
+  const int cw = c4 < nthread ? c4 : nthread;
+  const int rpb = nthread / cw;
+  const int tc = threadIdx.x % cw;
+  const int tr = threadIdx.x / cw;
+  const bool active = tr < rpb;
+
+  float4 s1[KQ], s2[KQ];
+  #pragma unroll
+  for (int k = 0; k < KQ; ++k) {
+    s1[k] = make_float4(0.f, 0.f, 0.f, 0.f);
+    s2[k] = make_float4(0.f, 0.f, 0.f, 0.f);
+  }
+  const float4* dy4 = reinterpret_cast<const float4*>(dy);
+  const float4* y4 = reinterpret_cast<const float4*>(y);
+  const float4* x4 = reinterpret_cast<const float4*>(x);
+  const float4* mean4 = reinterpret_cast<const float4*>(mean);
+  const float4* inv4 = reinterpret_cast<const float4*>(invstd);
+  const int64_t row_stride = (int64_t)rpb * 4 * gridDim.x;
+  if (active) {
+    for (int64_t row = (int64_t)blockIdx.x * rpb * 4 + tr; row < m;
+         row += row_stride) {
+      #pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int64_t r = row + (int64_t)j * rpb;
+        if (r >= m) continue;
+        const int64_t base = r * c4;
+        #pragma unroll
+        for (int k = 0; k < KQ; ++k) {
+          const int q = tc + k * cw;
+          if (q < c4) {
+            float4 g = dy4[base + q];
+            if (relu) {
+              const float4 yy = y4[base + q];
+              g.x = yy.x > 0.f ? g.x : 0.f;
+              g.y = yy.y > 0.f ? g.y : 0.f;
+              g.z = yy.z > 0.f ? g.z : 0.f;
+              g.w = yy.w > 0.f ? g.w : 0.f;
+            }
+            const float4 v = x4[base + q];
+            const float4 mu = mean4[q], is = inv4[q];
+            s1[k].x += g.x; s1[k].y += g.y;
+            s1[k].z += g.z; s1[k].w += g.w;
+            s2[k].x = fmaf(g.x, (v.x - mu.x) * is.x, s2[k].x);
+            s2[k].y = fmaf(g.y, (v.y - mu.y) * is.y, s2[k].y);
+            s2[k].z = fmaf(g.z, (v.z - mu.z) * is.z, s2[k].z);
+            s2[k].w = fmaf(g.w, (v.w - mu.w) * is.w, s2[k].w);
+          }
+        }
+      }
+    }
+  }
+
+  __shared__ Quad scratch[256];
+  const bool cw_pow2 = (cw & (cw - 1)) == 0;
+  const int rows_in_wave =
+      (cw_pow2 && cw < WAVE_SIZE) ? (WAVE_SIZE / cw) : 1;
+  #pragma unroll
+  for (int k = 0; k < KQ; ++k) {
+    float4 t1 = s1[k], t2 = s2[k];
+    for (int off = cw * (rows_in_wave >> 1); off >= cw && off > 0;
+         off >>= 1) {
+      t1.x += __shfl_down(t1.x, off, WAVE_SIZE);
+      t1.y += __shfl_down(t1.y, off, WAVE_SIZE);
+      t1.z += __shfl_down(t1.z, off, WAVE_SIZE);
+      t1.w += __shfl_down(t1.w, off, WAVE_SIZE);
+      t2.x += __shfl_down(t2.x, off, WAVE_SIZE);
+      t2.y += __shfl_down(t2.y, off, WAVE_SIZE);
+      t2.z += __shfl_down(t2.z, off, WAVE_SIZE);
+      t2.w += __shfl_down(t2.w, off, WAVE_SIZE);
+    }
+    scratch[threadIdx.x].s = t1;
+    scratch[threadIdx.x].ss = t2;
+    __syncthreads();
+    if (tr == 0) {
+      float4 f1 = t1, f2 = t2;
+      for (int r = rows_in_wave; r < rpb; r += rows_in_wave) {
+        const Quad& o = scratch[tc + r * cw];
+        f1.x += o.s.x; f1.y += o.s.y; f1.z += o.s.z; f1.w += o.s.w;
+        f2.x += o.ss.x; f2.y += o.ss.y; f2.z += o.ss.z; f2.w += o.ss.w;
+      }
+      const int q = tc + k * cw;
+      if (q < c4) {
+        const int ch = q * 4;
+        atomicAdd(&red_slot[ch + 0], f1.x);
+        atomicAdd(&red_slot[ch + 1], f1.y);
+        atomicAdd(&red_slot[ch + 2], f1.z);
+        atomicAdd(&red_slot[ch + 3], f1.w);
+        atomicAdd(&red_slot[c + ch + 0], f2.x);
+        atomicAdd(&red_slot[c + ch + 1], f2.y);
+        atomicAdd(&red_slot[c + ch + 2], f2.z);
+        atomicAdd(&red_slot[c + ch + 3], f2.w);
+      }
+    }
+    __syncthreads();
+  }
+}
+
+void launch_bn_bwd_reduce_v2(const float* dy, const float* y, const float* x,
+                             const float* mean, const float* invstd,
+                             float* red, int64_t m, int c, int relu,
+                             int slot_mask, int grid, hipStream_t stream) {
+  const dim3 g(grid), b(256);
+  switch (kq_for(c)) {
+    case 1: hipLaunchKernelGGL(bn_bwd_reduce_v2_kernel<1>, g, b, 0, stream,
+                               dy, y, x, mean, invstd, red, m, c, relu,
+                               slot_mask); break;
+    case 2: hipLaunchKernelGGL(bn_bwd_reduce_v2_kernel<2>, g, b, 0, stream,
+                               dy, y, x, mean, invstd, red, m, c, relu,
+                               slot_mask); break;
+    case 4: hipLaunchKernelGGL(bn_bwd_reduce_v2_kernel<4>, g, b, 0, stream,
+                               dy, y, x, mean, invstd, red, m, c, relu,
+                               slot_mask); break;
+    default: hipLaunchKernelGGL(bn_bwd_reduce_v2_kernel<8>, g, b, 0, stream,
+                                dy, y, x, mean, invstd, red, m, c, relu,
+                                slot_mask); break;
+  }
+}

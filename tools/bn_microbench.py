@@ -74,10 +74,40 @@ def main():
                        args.iters)
             if best[0] is None or t2 < best[0]:
                 best = (t2, grid)
+        # bwd_reduce v1 vs v2 on the same shape (3 input streams)
+        dy = torch.randn(m, c, device="cuda")
+        y = torch.randn(m, c, device="cuda")
+        mean = torch.zeros(c, device="cuda")
+        invstd = torch.ones(c, device="cuda")
+        red = torch.zeros(NSLOTS * 2 * c, device="cuda")
+        tb1 = bench(lambda: ext.bn_bwd_reduce(dy, y, x, mean, invstd, red,
+                                              m, c, 1, NSLOTS), args.iters)
+        refb = torch.zeros(2 * c, device="cuda")
+        red.zero_()
+        ext.bn_bwd_reduce(dy, y, x, mean, invstd, red, m, c, 1, NSLOTS)
+        ext.bn_reduce_slots(red, refb, NSLOTS)
+        bestb = (None, None)
+        for grid in (1024, 2048, 4096):
+            red.zero_()
+            ext.bn_bwd_reduce_v2(dy, y, x, mean, invstd, red, m, c, 1,
+                                 NSLOTS, grid)
+            gotb = torch.zeros(2 * c, device="cuda")
+            ext.bn_reduce_slots(red, gotb, NSLOTS)
+            assert torch.allclose(gotb, refb, rtol=1e-4, atol=1e-2), \
+                (m, c, grid, (gotb - refb).abs().max().item())
+            t2 = bench(lambda gd=grid: ext.bn_bwd_reduce_v2(
+                dy, y, x, mean, invstd, red, m, c, 1, NSLOTS, gd),
+                args.iters)
+            if bestb[0] is None or t2 < bestb[0]:
+                bestb = (t2, grid)
+        bwd_bytes = 3 * m * c * 4
         print(f"{m:>14} x {c:<5} {t1 * 1e3:>8.3f} "
               f"{bytes_read / t1 / 1e12:>8.2f} {best[0] * 1e3:>10.3f} "
-              f"{bytes_read / best[0] / 1e12:>6.2f} {best[1]:>6}")
-        del x, acc
+              f"{bytes_read / best[0] / 1e12:>6.2f} {best[1]:>6}  "
+              f"bwd {tb1 * 1e3:.3f}({bwd_bytes / tb1 / 1e12:.2f}TB/s) -> "
+              f"{bestb[0] * 1e3:.3f}({bwd_bytes / bestb[0] / 1e12:.2f}) "
+              f"g{bestb[1]}")
+        del x, acc, dy, y, red
 
 
 if __name__ == "__main__":
