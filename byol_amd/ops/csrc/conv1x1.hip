@@ -656,6 +656,123 @@ void conv1x1_wgrad_partial_kernel(const float* __restrict__ dY,
 #undef WG_X
 }
 
+// ---------------------------------------------------------------------------
+// wgrad v3: glds double-buffered m-step pipeline + the v1 atomic epilogue.
+// v1 stages synchronously (load 32 KB, barrier, compute, barrier — no
+// overlap); v2's partial slabs lost to v1 (the reduce pass costs more than
+// the atomics).  Both wgrad operands stage row-major from contiguous
+// global rows, so glds needs no source swizzle and the column reads are
+// conflict-free.  Requires M%32==0, N%TN==0, K%TK==0.
+// ---------------------------------------------------------------------------
+template <int TN, int TK, bool PRIO>
+__global__ __launch_bounds__(256)
+void conv1x1_wgrad_v3_kernel(const float* __restrict__ dY,
+                             const float* __restrict__ X,
+                             float* __restrict__ dW,
+                             int64_t M, int N, int K, int64_t chunk) {
+  constexpr int AT = TN / 64;  // 32x32 acc tiles per wave (n dim)
+  constexpr int AU = TK / 64;  //                          (k dim)
+  constexpr int LDSA = 32 * TN;
+  constexpr int LDSB = 32 * TK;
+  __shared__ __attribute__((aligned(16))) float lds[2 * (LDSA + LDSB)];
+
+  const int ntiles_k = K / TK;
+  const int ntiles_n = N / TN;
+  const int tile_n = (blockIdx.x / ntiles_k) % ntiles_n;
+  const int tile_k = blockIdx.x % ntiles_k;
+  const int64_t mchunk = blockIdx.x / (ntiles_k * ntiles_n);
+  const int n0 = tile_n * TN;
+  const int k0 = tile_k * TK;
+  const int64_t mstart = mchunk * chunk;
+  const int64_t mend = (mstart + chunk < M) ? mstart + chunk : M;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = (wave >> 1) * (TN / 2);
+  const int wc = (wave & 1) * (TK / 2);
+
+  f32x16 acc[AT][AU];
+  #pragma unroll
+  for (int t = 0; t < AT; ++t)
+    #pragma unroll
+    for (int u = 0; u < AU; ++u)
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) acc[t][u][e] = 0.f;
+
+  auto stage = [&](int buf, int64_t m0) {
+    #pragma unroll
+    for (int i = 0; i < TN / 32; ++i) {  // dY tile: [32][TN]
+      const int off = (wave * (TN / 32) + i) * 256 + lane * 4;
+      const int r = off / TN;
+      const int c = off % TN;
+      const float* src = dY + (m0 + r) * N + n0 + c;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * (LDSA + LDSB) + (wave * (TN / 32) + i) * 256],
+          16, 0, 0);
+    }
+    #pragma unroll
+    for (int i = 0; i < TK / 32; ++i) {  // X tile: [32][TK]
+      const int off = (wave * (TK / 32) + i) * 256 + lane * 4;
+      const int r = off / TK;
+      const int c = off % TK;
+      const float* src = X + (m0 + r) * K + k0 + c;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * (LDSA + LDSB) + LDSA
+                   + (wave * (TK / 32) + i) * 256],
+          16, 0, 0);
+    }
+  };
+
+  stage(0, mstart);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int buf = 0;
+  for (int64_t m0 = mstart; m0 < mend; m0 += 32) {
+    if (m0 + 32 < mend) stage(buf ^ 1, m0 + 32);
+    const float* ldy = &lds[buf * (LDSA + LDSB)];
+    const float* ldx = &lds[buf * (LDSA + LDSB) + LDSA];
+    if (PRIO) __builtin_amdgcn_s_setprio(1);
+    #pragma unroll
+    for (int kk = 0; kk < 32; kk += 2) {
+      const int mrow = kk + (lane >> 5);
+      #pragma unroll
+      for (int t = 0; t < AT; ++t) {
+        const float a = ldy[mrow * TN + wr + t * 32 + (lane & 31)];
+        #pragma unroll
+        for (int u = 0; u < AU; ++u) {
+          const float b = ldx[mrow * TK + wc + u * 32 + (lane & 31)];
+          acc[t][u] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc[t][u],
+                                                           0, 0, 0);
+        }
+      }
+    }
+    if (PRIO) __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+
+  #pragma unroll
+  for (int t = 0; t < AT; ++t) {
+    #pragma unroll
+    for (int u = 0; u < AU; ++u) {
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        const int row = (e & 3) + 8 * (e >> 2) + 4 * (lane >> 5);
+        const int col = lane & 31;
+        const int gn = n0 + wr + t * 32 + row;
+        const int gk = k0 + wc + u * 32 + col;
+        atomicAdd(&dW[(int64_t)gn * K + gk], acc[t][u][e]);
+      }
+    }
+  }
+}
+
 // dW[N*K] = sum over chunks of partial[chunk][N*K]
 __global__ void conv1x1_wgrad_reduce_kernel(const float* __restrict__ partial,
                                             float* __restrict__ dw,
@@ -730,8 +847,46 @@ void launch_conv1x1_wgrad_partial(const float* dy, const float* x,
                      dw, nk, nchunks);
 }
 
+static bool wgrad_v3_enabled() {
+  static bool f = [] {
+    const char* v = getenv("BYOL_WGRAD");
+    return v != nullptr && v[0] == 'v' && v[1] == '3';
+  }();
+  return f;
+}
+
+template <int TN, int TK>
+static void launch_wgrad_v3_tile(const float* dy, const float* x, float* dw,
+                                 int64_t m, int n, int k,
+                                 hipStream_t stream) {
+  const int64_t tiles = (int64_t)(n / TN) * (k / TK);
+  int64_t chunk = 32 * BK;
+  while (cdiv(m, chunk) * tiles > 4096) chunk *= 2;
+  const int64_t grid = tiles * cdiv(m, chunk);
+  if (conv_prio())
+    hipLaunchKernelGGL((conv1x1_wgrad_v3_kernel<TN, TK, true>),
+                       dim3((uint32_t)grid), dim3(256), 0, stream, dy, x,
+                       dw, m, n, k, chunk);
+  else
+    hipLaunchKernelGGL((conv1x1_wgrad_v3_kernel<TN, TK, false>),
+                       dim3((uint32_t)grid), dim3(256), 0, stream, dy, x,
+                       dw, m, n, k, chunk);
+}
+
 void launch_conv1x1_wgrad(const float* dy, const float* x, float* dw,
                           int64_t m, int n, int k, hipStream_t stream) {
+  if (wgrad_v3_enabled() && m % 32 == 0 && n % 64 == 0 && k % 64 == 0) {
+    const bool n128 = n % 128 == 0, k128 = k % 128 == 0;
+    if (n128 && k128)
+      launch_wgrad_v3_tile<128, 128>(dy, x, dw, m, n, k, stream);
+    else if (n128)
+      launch_wgrad_v3_tile<128, 64>(dy, x, dw, m, n, k, stream);
+    else if (k128)
+      launch_wgrad_v3_tile<64, 128>(dy, x, dw, m, n, k, stream);
+    else
+      launch_wgrad_v3_tile<64, 64>(dy, x, dw, m, n, k, stream);
+    return;
+  }
   // chunk M so total blocks ~<= 4096 per (n,k) tile-grid
   const int64_t tiles = cdiv(n, BM) * cdiv(k, BN);
   int64_t chunk = 32 * BK;  // 1024 rows minimum
