@@ -28,12 +28,24 @@ def mfma_conv_mode() -> str:
 
 
 # Shapes (K=Cin, N=Cout) where the MFMA FORWARD kernel measured faster
-# than MIOpen on MI355X at bs-512-class M (tools/conv_microbench.py,
-# profiles/r01_conv1x1_microbench.md); "auto" mode enables exactly these
-# and routes backward to MIOpen (aten.convolution_backward), which still
-# wins there.  BYOL_MFMA_CONV1X1=1 forces our kernels everywhere (tests).
+# than MIOpen on MI355X at bs-512-class M (tools/conv_microbench.py;
+# profiles/r02_validation.md — the 128x64 fast tile added the two N=64
+# shapes in round 2).  BYOL_MFMA_CONV1X1=1 forces our kernels everywhere
+# (tests).
 _AUTO_SHAPES = {
-    (64, 256), (256, 128), (128, 512), (512, 128), (512, 256), (256, 1024),
+    (64, 64), (64, 256), (256, 64), (256, 128),
+    (128, 512), (512, 128), (512, 256), (256, 1024),
+}
+
+# Shapes where OUR dgrad kernel beats MIOpen's split backward (r2 call 3:
+# 11 of 12 ResNet-50 shapes — e.g. 0.202 vs 0.238 ms on 56/64/64, 0.896
+# vs 1.015 on 28/512/256).  In auto mode backward runs our dgrad on these
+# + MIOpen wgrad (wgrad-only convolution_backward); MIOpen wrw still wins
+# wgrad everywhere (v3 pipeline pending measurement).
+_AUTO_DGRAD = {
+    (64, 64), (64, 256), (256, 64), (256, 128),
+    (128, 512), (512, 128), (512, 256), (256, 1024),
+    (1024, 256), (1024, 512), (2048, 512),
 }
 
 
@@ -50,7 +62,8 @@ def _eligible(x: torch.Tensor, weight: torch.Tensor) -> bool:
         return False
     if mode == "1":
         return True
-    return (k, weight.shape[0]) in _AUTO_SHAPES
+    shape = (k, weight.shape[0])
+    return shape in _AUTO_SHAPES or shape in _AUTO_DGRAD
 
 
 def _rows(t: torch.Tensor, c: int) -> torch.Tensor:
@@ -101,9 +114,17 @@ class _Conv1x1Fn(torch.autograd.Function):
             weight, "1x1",
             lambda: (lambda v: (v, v.t().contiguous()))(
                 weight.reshape(n, k).contiguous()))
-        y = torch.empty((b, n, h, w), device=x.device, dtype=x.dtype,
-                        memory_format=torch.channels_last)
-        C.conv1x1_fwd(_rows(x, k), wv, wt, _rows(y, n), m, k, n)
+        use_ours = mfma_conv_mode() == "1" or (k, n) in _AUTO_SHAPES
+        if use_ours:
+            y = torch.empty((b, n, h, w), device=x.device, dtype=x.dtype,
+                            memory_format=torch.channels_last)
+            C.conv1x1_fwd(_rows(x, k), wv, wt, _rows(y, n), m, k, n)
+        else:
+            # shape is here only for its backward win (_AUTO_DGRAD):
+            # MIOpen forward
+            y = torch.ops.aten.convolution(
+                x, wv.view(n, k, 1, 1), None, [1, 1], [0, 0], [1, 1],
+                False, [0, 0], 1)
         ctx.save_for_backward(x, wv)
         ctx.dims = (b, k, h, w, n, m)
         return y
@@ -130,8 +151,17 @@ class _Conv1x1Fn(torch.autograd.Function):
                 dw = torch.zeros(n, k, device=x.device, dtype=x.dtype)
                 C.conv1x1_wgrad(_rows(dy, n), _rows(x, k), dw, m, n, k)
             return dx, dw.view(n, k, 1, 1)
-        # auto mode: MIOpen backward still wins these shapes (see
-        # profiles/r01_conv1x1_microbench.md) — route dgrad/wgrad there
+        # auto mode: our dgrad where it measured faster (r2 call 3 —
+        # _AUTO_DGRAD), MIOpen for wgrad (its wrw igemm still wins) and
+        # for the rest
+        if (k, n) in _AUTO_DGRAD and m % 128 == 0:
+            C = extension()
+            dx = torch.empty_like(x)
+            C.conv1x1_dgrad(_rows(dy, n), wv, _rows(dx, k), m, n, k)
+            _, dw, _ = torch.ops.aten.convolution_backward(
+                dy, x, wv.view(n, k, 1, 1), [0], [1, 1], [0, 0], [1, 1],
+                False, [0, 0], 1, [False, True, False])
+            return dx, dw
         dx, dw, _ = torch.ops.aten.convolution_backward(
             dy, x, wv.view(n, k, 1, 1), [0], [1, 1], [0, 0], [1, 1], False,
             [0, 0], 1, [True, True, False])
