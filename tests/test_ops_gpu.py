@@ -620,3 +620,32 @@ def test_aug_blur_matches_composed(ext):
     got = out.permute(0, 3, 1, 2)
     assert torch.allclose(got, ref, rtol=1e-4, atol=1e-5), \
         (got - ref).abs().max().item()
+
+
+def test_auto_dispatch_backward_parity(ext, monkeypatch):
+    """Auto mode (per-shape tables): fwd+bwd parity vs F.conv2d on a
+    fwd-table shape, a dgrad-only shape, and an off-table shape."""
+    import torch.nn.functional as F
+    from byol_amd.ops.conv import _AUTO_DGRAD, _AUTO_SHAPES, _Conv1x1Fn
+    monkeypatch.setenv("BYOL_MFMA_CONV1X1", "auto")
+    torch.manual_seed(27)
+    cases = [(8, 56, 64, 256),     # fwd + dgrad tables
+             (32, 14, 1024, 256),  # dgrad-only table (MIOpen fwd)
+             (32, 7, 512, 2048)]   # off both tables
+    for bsz, hgt, cin, cout in cases:
+        x = torch.randn(bsz, cin, hgt, hgt, device="cuda").to(
+            memory_format=torch.channels_last).requires_grad_(True)
+        w = (torch.randn(cout, cin, 1, 1, device="cuda") * 0.05
+             ).requires_grad_(True)
+        y = _Conv1x1Fn.apply(x, w)
+        g = torch.randn_like(y).to(memory_format=torch.channels_last)
+        y.backward(g)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        F.conv2d(x2, w2).backward(g)
+        assert torch.allclose(y, F.conv2d(x.detach(), w.detach()),
+                              rtol=1e-4, atol=1e-4)
+        assert torch.allclose(x.grad, x2.grad, rtol=1e-4, atol=1e-4), \
+            (cin, cout, (x.grad - x2.grad).abs().max().item())
+        assert torch.allclose(w.grad, w2.grad, rtol=1e-3, atol=1e-2)
+    assert (64, 64) in _AUTO_SHAPES and (1024, 256) in _AUTO_DGRAD
