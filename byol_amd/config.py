@@ -112,6 +112,12 @@ def build_parser() -> argparse.ArgumentParser:
     parser.add_argument("--half", action="store_true", default=False,
                         help="half precision training (bf16 autocast on "
                              "MI355X; the reference used Apex fp16 O2)")
+    parser.add_argument("--half-dtype", type=str, default="bf16",
+                        choices=["bf16", "fp16"],
+                        help="--half compute dtype: bf16 (MI355X-native, no "
+                             "loss scaler needed) or fp16 + dynamic "
+                             "GradScaler for strict parity with the "
+                             "reference's Apex O2 fp16 experiments")
 
     # MI355X-native knobs (not in the reference)
     parser.add_argument("--bucket-cap-mb", type=float, default=32.0,

@@ -196,8 +196,15 @@ def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
     loss_map, num_samples = {}, 0
     num_minibatches = 0
     is_ddp = isinstance(model, FlatDDP)
-    autocast_ctx = (torch.autocast("cuda", dtype=torch.bfloat16)
-                    if args.half and args.cuda else _dummy_context())
+    # --half: bf16 autocast by default (no loss scaler needed on MI355X);
+    # --half-dtype fp16 adds a dynamic GradScaler for strict parity with
+    # the reference's Apex O2 fp16 runs (/root/reference/main.py:614,746)
+    use_fp16 = args.half and getattr(args, "half_dtype", "bf16") == "fp16"
+    autocast_ctx = (torch.autocast(
+        "cuda", dtype=torch.float16 if use_fp16 else torch.bfloat16)
+        if args.half and args.cuda else _dummy_context())
+    scaler = torch.amp.GradScaler(
+        "cuda", enabled=use_fp16 and args.cuda and not is_eval)
     from ..profiling import PhaseTimer
     timer = PhaseTimer(enabled=getattr(args, "perf_stats", False),
                        use_cuda=args.cuda)
@@ -222,12 +229,22 @@ def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
                 torch.cat([lab, lab], 0))
             total = byol_loss + classifier_loss
         optimizer.zero_grad()
-        total.backward()
-        if is_ddp:
-            model.finish_grad_sync()
-        if args.clip > 0:
-            nn.utils.clip_grad_value_(model.parameters(), args.clip)
-        optimizer.step()
+        if scaler.is_enabled():
+            scaler.scale(total).backward()
+            if is_ddp:
+                model.finish_grad_sync()
+            scaler.unscale_(optimizer)
+            if args.clip > 0:
+                nn.utils.clip_grad_value_(model.parameters(), args.clip)
+            scaler.step(optimizer)
+            scaler.update()
+        else:
+            total.backward()
+            if is_ddp:
+                model.finish_grad_sync()
+            if args.clip > 0:
+                nn.utils.clip_grad_value_(model.parameters(), args.clip)
+            optimizer.step()
         if args.polyak_ema > 0:
             layers.polyak_ema_parameters(model, args.polyak_ema)
         return total, byol_loss, classifier_loss, acc1, acc5
@@ -236,9 +253,12 @@ def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
     # owns the EMA decay/LR device scalars; DDP+graph is a later
     # experiment).  First GRAPH_WARM minibatches run eager (MIOpen finds,
     # fused-LARS state), then the step is captured once and replayed.
+    # (fp16 GradScaler does host-side inf checks each step — not
+    # graph-capturable, so fp16 excludes the graph path)
     graph_wanted = (not is_eval and args.cuda
                     and getattr(args, "hip_graph", False)
-                    and args.polyak_ema == 0 and not is_ddp)
+                    and args.polyak_ema == 0 and not is_ddp
+                    and not scaler.is_enabled())
     GRAPH_WARM = 2
     graphed = None
 

@@ -649,3 +649,39 @@ def test_auto_dispatch_backward_parity(ext, monkeypatch):
             (cin, cout, (x.grad - x2.grad).abs().max().item())
         assert torch.allclose(w.grad, w2.grad, rtol=1e-3, atol=1e-2)
     assert (64, 64) in _AUTO_SHAPES and (1024, 256) in _AUTO_DGRAD
+
+
+def test_fp16_scaler_training_step(ext):
+    """--half-dtype fp16 parity mode: autocast fp16 + GradScaler over the
+    fused-LARS step trains without inf/nan for a few steps."""
+    from byol_amd.objective import loss_function
+    from byol_amd.ops.classifier import cross_entropy_topk
+
+    m = _tiny_byol(seed=55)
+    opt = _lars_for(m)
+    scaler = torch.amp.GradScaler("cuda")
+    B, S = 16, 32
+    for i in range(3):
+        a1 = torch.rand(B, 3, S, S, device="cuda").to(
+            memory_format=torch.channels_last)
+        a2 = torch.rand(B, 3, S, S, device="cuda").to(
+            memory_format=torch.channels_last)
+        lab = torch.randint(10, (B,), device="cuda")
+        with torch.autocast("cuda", dtype=torch.float16):
+            out = m(a1, a2)
+            loss = loss_function(
+                out["online_prediction1"].float(),
+                out["online_prediction2"].float(),
+                out["target_projection1"].float(),
+                out["target_projection2"].float())
+            ce, _1, _5 = cross_entropy_topk(
+                out["linear_preds"].float().contiguous(),
+                torch.cat([lab, lab]))
+            loss = loss + ce
+        opt.zero_grad()
+        scaler.scale(loss).backward()
+        scaler.unscale_(opt)
+        scaler.step(opt)
+        scaler.update()
+    assert torch.isfinite(m.flat_space.flat_params).all()
+    assert torch.isfinite(loss.float())
