@@ -40,12 +40,18 @@ _AUTO_SHAPES = {
 # Shapes where OUR dgrad kernel beats MIOpen's split backward (r2 call 3:
 # 11 of 12 ResNet-50 shapes — e.g. 0.202 vs 0.238 ms on 56/64/64, 0.896
 # vs 1.015 on 28/512/256).  In auto mode backward runs our dgrad on these
-# + MIOpen wgrad (wgrad-only convolution_backward); MIOpen wrw still wins
-# wgrad everywhere (v3 pipeline pending measurement).
+# + per-shape wgrad routing below.
 _AUTO_DGRAD = {
     (64, 64), (64, 256), (256, 64), (256, 128),
     (128, 512), (512, 128), (512, 256), (256, 1024),
     (1024, 256), (1024, 512), (2048, 512),
+}
+
+# Shapes where OUR wgrad (v3 glds pipeline) beats MIOpen wrw (r2 call 4:
+# 0.536 vs 0.559 on 56/64/256, 0.540 vs 0.558 on 56/256/64, 0.912 vs
+# 0.938 on 56/256/128); the rest stay with MIOpen (3-8% faster there).
+_AUTO_WGRAD = {
+    (64, 256), (256, 64), (256, 128),
 }
 
 
@@ -151,16 +157,28 @@ class _Conv1x1Fn(torch.autograd.Function):
                 dw = torch.zeros(n, k, device=x.device, dtype=x.dtype)
                 C.conv1x1_wgrad(_rows(dy, n), _rows(x, k), dw, m, n, k)
             return dx, dw.view(n, k, 1, 1)
-        # auto mode: our dgrad where it measured faster (r2 call 3 —
-        # _AUTO_DGRAD), MIOpen for wgrad (its wrw igemm still wins) and
-        # for the rest
-        if (k, n) in _AUTO_DGRAD and m % 128 == 0:
+        # auto mode: per-shape measured routing (r2 calls 3/4) — our dgrad
+        # on _AUTO_DGRAD, our wgrad v3 on _AUTO_WGRAD, MIOpen elsewhere
+        use_dgrad = (k, n) in _AUTO_DGRAD and m % 128 == 0
+        use_wgrad = ((k, n) in _AUTO_WGRAD and m % 32 == 0
+                     and n % 64 == 0 and k % 64 == 0)
+        if use_dgrad or use_wgrad:
             C = extension()
-            dx = torch.empty_like(x)
-            C.conv1x1_dgrad(_rows(dy, n), wv, _rows(dx, k), m, n, k)
-            _, dw, _ = torch.ops.aten.convolution_backward(
-                dy, x, wv.view(n, k, 1, 1), [0], [1, 1], [0, 0], [1, 1],
-                False, [0, 0], 1, [False, True, False])
+            if use_dgrad:
+                dx = torch.empty_like(x)
+                C.conv1x1_dgrad(_rows(dy, n), wv, _rows(dx, k), m, n, k)
+            else:
+                dx, _, _ = torch.ops.aten.convolution_backward(
+                    dy, x, wv.view(n, k, 1, 1), [0], [1, 1], [0, 0],
+                    [1, 1], False, [0, 0], 1, [True, False, False])
+            if use_wgrad:
+                dw = torch.zeros(n, k, device=x.device, dtype=x.dtype)
+                C.conv1x1_wgrad(_rows(dy, n), _rows(x, k), dw, m, n, k)
+                dw = dw.view(n, k, 1, 1)
+            else:
+                _, dw, _ = torch.ops.aten.convolution_backward(
+                    dy, x, wv.view(n, k, 1, 1), [0], [1, 1], [0, 0],
+                    [1, 1], False, [0, 0], 1, [False, True, False])
             return dx, dw
         dx, dw, _ = torch.ops.aten.convolution_backward(
             dy, x, wv.view(n, k, 1, 1), [0], [1, 1], [0, 0], [1, 1], False,

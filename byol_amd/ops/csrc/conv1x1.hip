@@ -848,9 +848,12 @@ void launch_conv1x1_wgrad_partial(const float* dy, const float* x,
 }
 
 static bool wgrad_v3_enabled() {
+  // v3 (glds pipeline + right-sized tiles) measured 2-4x v1 on every
+  // ResNet-50 shape (profiles/r02_validation.md call 4) — default ON for
+  // eligible geometry; BYOL_WGRAD=atomic forces the v1 kernel for A/Bs.
   static bool f = [] {
     const char* v = getenv("BYOL_WGRAD");
-    return v != nullptr && v[0] == 'v' && v[1] == '3';
+    return v == nullptr || !(v[0] == 'a' || v[0] == '1');
   }();
   return f;
 }
