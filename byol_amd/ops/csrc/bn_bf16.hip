@@ -62,12 +62,14 @@ __global__ void bn_stats_bf16_kernel(const ushort* __restrict__ x,
     ss[k] = make_float4(0.f, 0.f, 0.f, 0.f);
   }
 
-  const int64_t row_stride = (int64_t)rpb * 4 * gridDim.x;
+  // 8-row unroll + shfl tail: the structure the fp32 v2 sweep won with
+  // (r2 call 4) ported to bf16 I/O
+  const int64_t row_stride = (int64_t)rpb * 8 * gridDim.x;
   if (active) {
-    for (int64_t row = (int64_t)blockIdx.x * rpb * 4 + tr; row < m;
+    for (int64_t row = (int64_t)blockIdx.x * rpb * 8 + tr; row < m;
          row += row_stride) {
       #pragma unroll
-      for (int j = 0; j < 4; ++j) {
+      for (int j = 0; j < 8; ++j) {
         const int64_t r = row + (int64_t)j * rpb;
         if (r < m) {
           const int64_t base = r * c4;
@@ -89,14 +91,26 @@ __global__ void bn_stats_bf16_kernel(const ushort* __restrict__ x,
   }
 
   __shared__ QuadBF scratch[256];
+  const bool cw_pow2 = (cw & (cw - 1)) == 0;
+  const int riw = (cw_pow2 && cw < 64) ? (64 / cw) : 1;
   #pragma unroll
   for (int k = 0; k < KQ; ++k) {
-    scratch[threadIdx.x].s = s[k];
-    scratch[threadIdx.x].ss = ss[k];
+    float4 ts = s[k], tss = ss[k];
+    for (int off = cw * (riw >> 1); off >= cw && off > 0; off >>= 1) {
+      ts.x += __shfl_down(ts.x, off, 64);
+      ts.y += __shfl_down(ts.y, off, 64);
+      ts.z += __shfl_down(ts.z, off, 64);
+      ts.w += __shfl_down(ts.w, off, 64);
+      tss.x += __shfl_down(tss.x, off, 64);
+      tss.y += __shfl_down(tss.y, off, 64);
+      tss.z += __shfl_down(tss.z, off, 64);
+      tss.w += __shfl_down(tss.w, off, 64);
+    }
+    scratch[threadIdx.x].s = ts;
+    scratch[threadIdx.x].ss = tss;
     __syncthreads();
     if (tr == 0) {
-      float4 ts = s[k], tss = ss[k];
-      for (int r = 1; r < rpb; ++r) {
+      for (int r = riw; r < rpb; r += riw) {
         const QuadBF& o = scratch[tc + r * cw];
         ts.x += o.s.x; ts.y += o.s.y; ts.z += o.s.z; ts.w += o.s.w;
         tss.x += o.ss.x; tss.y += o.ss.y; tss.z += o.ss.z; tss.w += o.ss.w;
@@ -182,12 +196,12 @@ __global__ void bn_bwd_reduce_bf16_kernel(const ushort* __restrict__ dy,
   }
   const float4* mean4 = reinterpret_cast<const float4*>(mean);
   const float4* inv4 = reinterpret_cast<const float4*>(invstd);
-  const int64_t row_stride = (int64_t)rpb * 2 * gridDim.x;
+  const int64_t row_stride = (int64_t)rpb * 4 * gridDim.x;
   if (active) {
-    for (int64_t row = (int64_t)blockIdx.x * rpb * 2 + tr; row < m;
+    for (int64_t row = (int64_t)blockIdx.x * rpb * 4 + tr; row < m;
          row += row_stride) {
       #pragma unroll
-      for (int j = 0; j < 2; ++j) {
+      for (int j = 0; j < 4; ++j) {
         const int64_t r = row + (int64_t)j * rpb;
         if (r >= m) continue;
         const int64_t base = r * c4;
@@ -217,14 +231,26 @@ __global__ void bn_bwd_reduce_bf16_kernel(const ushort* __restrict__ dy,
   }
 
   __shared__ QuadBF scratch[256];
+  const bool cw_pow2 = (cw & (cw - 1)) == 0;
+  const int riw = (cw_pow2 && cw < 64) ? (64 / cw) : 1;
   #pragma unroll
   for (int k = 0; k < KQ; ++k) {
-    scratch[threadIdx.x].s = s1[k];
-    scratch[threadIdx.x].ss = s2[k];
+    float4 t1 = s1[k], t2 = s2[k];
+    for (int off = cw * (riw >> 1); off >= cw && off > 0; off >>= 1) {
+      t1.x += __shfl_down(t1.x, off, 64);
+      t1.y += __shfl_down(t1.y, off, 64);
+      t1.z += __shfl_down(t1.z, off, 64);
+      t1.w += __shfl_down(t1.w, off, 64);
+      t2.x += __shfl_down(t2.x, off, 64);
+      t2.y += __shfl_down(t2.y, off, 64);
+      t2.z += __shfl_down(t2.z, off, 64);
+      t2.w += __shfl_down(t2.w, off, 64);
+    }
+    scratch[threadIdx.x].s = t1;
+    scratch[threadIdx.x].ss = t2;
     __syncthreads();
     if (tr == 0) {
-      float4 t1 = s1[k], t2 = s2[k];
-      for (int r = 1; r < rpb; ++r) {
+      for (int r = riw; r < rpb; r += riw) {
         const QuadBF& o = scratch[tc + r * cw];
         t1.x += o.s.x; t1.y += o.s.y; t1.z += o.s.z; t1.w += o.s.w;
         t2.x += o.ss.x; t2.y += o.ss.y; t2.z += o.ss.z; t2.w += o.ss.w;
@@ -317,7 +343,8 @@ static int kq_for_bf(int c) {
 
 void launch_bn_stats_bf16(const ushort* x, float* acc, int64_t m, int c,
                           int slot_mask, hipStream_t stream) {
-  const dim3 g(stats_grid_bf(m, c)), b(256);
+  // grid knee from the fp32 v2 sweep (r2 call 4)
+  const dim3 g(m > 4000000 ? 2048 : 1024), b(256);
   switch (kq_for_bf(c)) {
     case 1: hipLaunchKernelGGL(bn_stats_bf16_kernel<1>, g, b, 0, stream, x,
                                acc, m, c, slot_mask); break;
@@ -345,7 +372,7 @@ void launch_bn_bwd_reduce_bf16(const ushort* dy, const ushort* y,
                                const float* invstd, float* red, int64_t m,
                                int c, int relu, int slot_mask,
                                hipStream_t stream) {
-  const dim3 g(stats_grid_bf(m, c)), b(256);
+  const dim3 g(m > 4000000 ? 2048 : 1024), b(256);
   switch (kq_for_bf(c)) {
     case 1: hipLaunchKernelGGL(bn_bwd_reduce_bf16_kernel<1>, g, b, 0, stream,
                                dy, y, x, mean, invstd, red, m, c, relu,
