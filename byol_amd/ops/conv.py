@@ -244,6 +244,28 @@ class _Conv3x3Fn(torch.autograd.Function):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous(memory_format=torch.channels_last)
         s = ctx.stride
+        b, k, hi, wi = x.shape
+        n = weight.shape[0]
+        ho = (hi + 2 - 3) // s + 1
+        wo = (wi + 2 - 3) // s + 1
+        m = b * ho * wo
+        # our 9-tap wgrad (v3 m-pipeline per tap, padded input) — gated by
+        # BYOL_C3WGRAD until the microbench proves per-shape wins
+        use_wgrad = (os.environ.get("BYOL_C3WGRAD", "0") == "1"
+                     and m % 32 == 0 and n % 64 == 0 and k % 64 == 0)
+        if use_wgrad:
+            C = extension()
+            dx, _, _ = torch.ops.aten.convolution_backward(
+                dy, x, weight, [0], [s, s], [1, 1], [1, 1], False, [0, 0],
+                1, [True, False, False])
+            xpad = torch.empty(b * (hi + 2) * (wi + 2) * k,
+                               device=x.device, dtype=x.dtype)
+            C.pad_nhwc(_rows(x, k), xpad, b, hi, wi, k)
+            dw9 = torch.zeros(9 * n * k, device=x.device, dtype=x.dtype)
+            dw = torch.empty(n, k, 3, 3, device=x.device, dtype=x.dtype)
+            C.conv3x3_wgrad(_rows(dy, n), xpad, dw9, dw.reshape(-1), b, hi,
+                            wi, ho, wo, k, n, s)
+            return dx, dw, None
         dx, dw, _ = torch.ops.aten.convolution_backward(
             dy, x, weight, [0], [s, s], [1, 1], [1, 1], False, [0, 0], 1,
             [True, True, False])

@@ -95,6 +95,9 @@ void launch_pad_nhwc(const float* x, float* xp, int b, int hi, int wi,
 void launch_conv3x3_fwd_fast(const float* xp, const float* wp, float* y,
                              int b, int hi, int wi, int ho, int wo, int k,
                              int n, int stride, hipStream_t stream);
+void launch_conv3x3_wgrad(const float* dy, const float* xp, float* dw9,
+                          float* dw, int b, int hi, int wi, int ho, int wo,
+                          int k, int n, int stride, hipStream_t stream);
 void launch_lars_momentum_step(float* p, const float* g, float* m,
                                float* norm_acc, float* alr,
                                const int64_t* seg_off,
@@ -172,6 +175,20 @@ std::tuple<torch::Tensor, torch::Tensor> byol_loss_backward(
       grad_out.contiguous().data_ptr<float>(), g1.data_ptr<float>(),
       g2.data_ptr<float>(), n, batch, stream);
   return {g1, g2};
+}
+
+void conv3x3_wgrad(torch::Tensor dy, torch::Tensor xp, torch::Tensor dw9,
+                   torch::Tensor dw, int64_t b, int64_t hi, int64_t wi,
+                   int64_t ho, int64_t wo, int64_t k, int64_t n,
+                   int64_t stride) {
+  CHECK_IN(dy); CHECK_IN(xp); CHECK_IN(dw9); CHECK_IN(dw);
+  TORCH_CHECK(dw9.numel() == 9 * n * k, "dw9 must be [9][N][K]");
+  TORCH_CHECK(dw.numel() == n * k * 9, "dw must be [N][K][3][3]");
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_conv3x3_wgrad(dy.data_ptr<float>(), xp.data_ptr<float>(),
+                       dw9.data_ptr<float>(), dw.data_ptr<float>(), (int)b,
+                       (int)hi, (int)wi, (int)ho, (int)wo, (int)k, (int)n,
+                       (int)stride, stream);
 }
 
 void lars_momentum_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
@@ -504,6 +521,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_bwd_apply_bf16", &bn_bwd_apply_bf16);
   mod.def("conv3x3_fwd", &conv3x3_fwd);
   mod.def("conv3x3_fwd_fast", &conv3x3_fwd_fast);
+  mod.def("conv3x3_wgrad", &conv3x3_wgrad);
   mod.def("pad_nhwc", &pad_nhwc);
   mod.def("conv1x1_fwd", &conv1x1_fwd);
   mod.def("conv1x1_dgrad", &conv1x1_dgrad);

@@ -480,3 +480,181 @@ void launch_conv3x3_fwd(const float* x, const float* wp, float* y, int b,
   hipLaunchKernelGGL(conv3x3_fwd_kernel, dim3((uint32_t)grid), dim3(256), 0,
                      stream, x, wp, y, b, hi, wi, ho, wo, k, n, stride);
 }
+
+// ---------------------------------------------------------------------------
+// 3x3 wgrad: dW[N,K,3,3] = sum_m dY[m,n] * Xpad(tap)[m,k] — nine per-tap
+// K-reduction GEMMs sharing the conv1x1 wgrad-v3 structure (glds
+// double-buffered m-step pipeline + atomic epilogue).  tap lives in the
+// GRID (each block owns one (n-tile, k-tile, tap, m-chunk)); the padded
+// input makes every tap an unguarded uniform shift.  Output goes to a
+// [9][N][K] workspace, permuted to torch's [N][K][3][3] by a tiny kernel.
+// Requires M%32==0, N%64==0, K%64==0.
+// ---------------------------------------------------------------------------
+template <int TN, int TK>
+__global__ __launch_bounds__(256)
+void conv3x3_wgrad_kernel(const float* __restrict__ dY,
+                          const float* __restrict__ Xp,  // padded NHWC
+                          float* __restrict__ dW9,       // [9][N][K]
+                          int B, int Hp, int Wpp, int Ho, int Wo,
+                          int K, int N, int stride, int64_t chunk) {
+  constexpr int AT = TN / 64;
+  constexpr int AU = TK / 64;
+  constexpr int LDSY = 32 * TN;
+  constexpr int LDSX = 32 * TK;
+  __shared__ __attribute__((aligned(16))) float lds[2 * (LDSY + LDSX)];
+
+  const int64_t M = (int64_t)B * Ho * Wo;
+  const int ntiles_k = K / TK;
+  const int ntiles_n = N / TN;
+  const int tile_k = blockIdx.x % ntiles_k;
+  const int tile_n = (blockIdx.x / ntiles_k) % ntiles_n;
+  const int tap = (blockIdx.x / (ntiles_k * ntiles_n)) % 9;
+  const int64_t mchunk = blockIdx.x / (ntiles_k * ntiles_n * 9);
+  const int n0 = tile_n * TN;
+  const int k0 = tile_k * TK;
+  const int dy_t = tap / 3, dx_t = tap % 3;
+  const int64_t mstart = mchunk * chunk;
+  const int64_t mend = (mstart + chunk < M) ? mstart + chunk : M;
+  if (mstart >= M) return;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = (wave >> 1) * (TN / 2);
+  const int wc = (wave & 1) * (TK / 2);
+  const int64_t howo = (int64_t)Ho * Wo;
+
+  f32x16_c3 acc[AT][AU];
+  #pragma unroll
+  for (int t = 0; t < AT; ++t)
+    #pragma unroll
+    for (int u = 0; u < AU; ++u)
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) acc[t][u][e] = 0.f;
+
+  auto stage = [&](int buf, int64_t m0) {
+    #pragma unroll
+    for (int i = 0; i < TN / 32; ++i) {  // dY tile [32][TN]
+      const int off = (wave * (TN / 32) + i) * 256 + lane * 4;
+      const int r = off / TN;
+      const int c = off % TN;
+      const float* src = dY + (m0 + r) * N + n0 + c;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * (LDSY + LDSX) + (wave * (TN / 32) + i) * 256],
+          16, 0, 0);
+    }
+    #pragma unroll
+    for (int i = 0; i < TK / 32; ++i) {  // Xp(tap) tile [32][TK]
+      const int off = (wave * (TK / 32) + i) * 256 + lane * 4;
+      const int r = off / TK;
+      const int c = off % TK;
+      const int64_t gm = m0 + r;
+      const int bb = (int)(gm / howo);
+      const int rem = (int)(gm % howo);
+      const int iy = (rem / Wo) * stride + dy_t;
+      const int ix = (rem % Wo) * stride + dx_t;
+      const float* src =
+          Xp + (((int64_t)bb * Hp + iy) * Wpp + ix) * K + k0 + c;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * (LDSY + LDSX) + LDSY
+                   + (wave * (TK / 32) + i) * 256],
+          16, 0, 0);
+    }
+  };
+
+  stage(0, mstart);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int buf = 0;
+  for (int64_t m0 = mstart; m0 < mend; m0 += 32) {
+    if (m0 + 32 < mend) stage(buf ^ 1, m0 + 32);
+    const float* ldy = &lds[buf * (LDSY + LDSX)];
+    const float* ldx = &lds[buf * (LDSY + LDSX) + LDSY];
+    #pragma unroll
+    for (int kk = 0; kk < 32; kk += 2) {
+      const int mrow = kk + (lane >> 5);
+      #pragma unroll
+      for (int t = 0; t < AT; ++t) {
+        const float a = ldy[mrow * TN + wr + t * 32 + (lane & 31)];
+        #pragma unroll
+        for (int u = 0; u < AU; ++u) {
+          const float b = ldx[mrow * TK + wc + u * 32 + (lane & 31)];
+          acc[t][u] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc[t][u],
+                                                           0, 0, 0);
+        }
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+
+  float* slab = dW9 + (int64_t)tap * N * K;
+  #pragma unroll
+  for (int t = 0; t < AT; ++t) {
+    #pragma unroll
+    for (int u = 0; u < AU; ++u) {
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        const int row = (e & 3) + 8 * (e >> 2) + 4 * (lane >> 5);
+        const int col = lane & 31;
+        const int gn = n0 + wr + t * 32 + row;
+        const int gk = k0 + wc + u * 32 + col;
+        atomicAdd(&slab[(int64_t)gn * K + gk], acc[t][u][e]);
+      }
+    }
+  }
+}
+
+// [9][N][K] workspace -> torch [N][K][3][3]
+__global__ void wgrad9_permute_kernel(const float* __restrict__ dW9,
+                                      float* __restrict__ dw,
+                                      int64_t nk) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nk;
+       i += stride) {
+    #pragma unroll
+    for (int tap = 0; tap < 9; ++tap)
+      dw[i * 9 + tap] = dW9[(int64_t)tap * nk + i];
+  }
+}
+
+template <int TN, int TK>
+static void launch_c3wg_tile(const float* dy, const float* xp, float* dw9,
+                             int b, int hp, int wpp, int ho, int wo, int k,
+                             int n, int stride, hipStream_t stream) {
+  const int64_t m = (int64_t)b * ho * wo;
+  const int64_t tiles = (int64_t)(n / TN) * (k / TK) * 9;
+  int64_t chunk = 1024;
+  while ((m + chunk - 1) / chunk * tiles > 8192) chunk *= 2;
+  const int64_t grid = tiles * ((m + chunk - 1) / chunk);
+  hipLaunchKernelGGL((conv3x3_wgrad_kernel<TN, TK>), dim3((uint32_t)grid),
+                     dim3(256), 0, stream, dy, xp, dw9, b, hp, wpp, ho, wo,
+                     k, n, stride, chunk);
+}
+
+void launch_conv3x3_wgrad(const float* dy, const float* xp, float* dw9,
+                          float* dw, int b, int hi, int wi, int ho, int wo,
+                          int k, int n, int stride, hipStream_t stream) {
+  const int hp = hi + 2, wpp = wi + 2;
+  const bool n128 = n % 128 == 0, k128 = k % 128 == 0;
+  if (n128 && k128)
+    launch_c3wg_tile<128, 128>(dy, xp, dw9, b, hp, wpp, ho, wo, k, n,
+                               stride, stream);
+  else if (n128)
+    launch_c3wg_tile<128, 64>(dy, xp, dw9, b, hp, wpp, ho, wo, k, n,
+                              stride, stream);
+  else if (k128)
+    launch_c3wg_tile<64, 128>(dy, xp, dw9, b, hp, wpp, ho, wo, k, n,
+                              stride, stream);
+  else
+    launch_c3wg_tile<64, 64>(dy, xp, dw9, b, hp, wpp, ho, wo, k, n,
+                             stride, stream);
+  const int64_t nk = (int64_t)n * k;
+  hipLaunchKernelGGL(wgrad9_permute_kernel, dim3(grid_1d(nk, 256)),
+                     dim3(256), 0, stream, dw9, dw, nk);
+}

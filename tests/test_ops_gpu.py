@@ -685,3 +685,31 @@ def test_fp16_scaler_training_step(ext):
         scaler.update()
     assert torch.isfinite(m.flat_space.flat_params).all()
     assert torch.isfinite(loss.float())
+
+
+def test_conv3x3_wgrad_matches_oracle(ext, monkeypatch):
+    """9-tap wgrad kernel (BYOL_C3WGRAD) vs MIOpen weight grads, stride
+    1 and 2."""
+    import torch.nn.functional as F
+    from byol_amd.ops.conv import _Conv3x3Fn
+    monkeypatch.setenv("BYOL_MFMA_CONV3X3", "1")
+    monkeypatch.setenv("BYOL_C3WGRAD", "1")
+    torch.manual_seed(31)
+    for bsz, hgt, c, stride in [(8, 16, 128, 1), (8, 16, 256, 2),
+                                (2, 32, 64, 1)]:
+        x = torch.randn(bsz, c, hgt, hgt, device="cuda").to(
+            memory_format=torch.channels_last).requires_grad_(True)
+        w = (torch.randn(c, c, 3, 3, device="cuda") * 0.05
+             ).requires_grad_(True)
+        ho = (hgt + 2 - 3) // stride + 1
+        if (bsz * ho * ho) % 32 != 0:
+            continue
+        y = _Conv3x3Fn.apply(x, w, stride)
+        g = torch.randn_like(y).to(memory_format=torch.channels_last)
+        y.backward(g)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        F.conv2d(x2, w2, stride=stride, padding=1).backward(g)
+        assert torch.allclose(x.grad, x2.grad, rtol=1e-4, atol=1e-4)
+        assert torch.allclose(w.grad, w2.grad, rtol=1e-3, atol=1e-2), \
+            (hgt, c, stride, (w.grad - w2.grad).abs().max().item())

@@ -125,8 +125,11 @@ def main():
 
 
 def main_3x3(batch=512, iters=8):
-    from byol_amd.ops.conv import _Conv3x3Fn
-    print(f"{'3x3 shape':>22} {'miopen f':>9} {'mfma f':>9}")
+    from byol_amd.ops import require_extension
+    from byol_amd.ops.conv import _Conv3x3Fn, _rows
+    C = require_extension("3x3 bench")
+    print(f"{'3x3 shape':>22} {'miopen f':>9} {'mfma f':>9} "
+          f"{'wg ours':>8} {'wg miopen':>9}")
     for hgt, c, stride in SHAPES_3X3:
         x = torch.randn(batch, c, hgt, hgt, device="cuda").to(
             memory_format=torch.channels_last)
@@ -135,9 +138,32 @@ def main_3x3(batch=512, iters=8):
         to = timeit(lambda: _Conv3x3Fn.apply(x, w, stride), iters)
         err = (F.conv2d(x, w, stride=stride, padding=1)
                - _Conv3x3Fn.apply(x, w, stride)).abs().max().item()
+        # 3x3 wgrad A/B: ours (9-tap v3) vs MIOpen wrw-only
+        ho = (hgt + 2 - 3) // stride + 1
+        y = F.conv2d(x, w, stride=stride, padding=1)
+        g = torch.randn_like(y).to(memory_format=torch.channels_last)
+        b = batch
+        xpad = torch.empty(b * (hgt + 2) * (hgt + 2) * c, device="cuda")
+        C.pad_nhwc(_rows(x, c), xpad, b, hgt, hgt, c)
+        dw9 = torch.zeros(9 * c * c, device="cuda")
+        dw = torch.empty(c, c, 3, 3, device="cuda")
+        def wg_ours():
+            dw9.zero_()
+            C.conv3x3_wgrad(_rows(g, c), xpad, dw9, dw.reshape(-1), b, hgt,
+                            hgt, ho, ho, c, c, stride)
+        t_wg = timeit(wg_ours, iters)
+        t_wg_m = timeit(lambda: torch.ops.aten.convolution_backward(
+            g, x, w, [0], [stride, stride], [1, 1], [1, 1], False, [0, 0],
+            1, [False, True, False]), iters)
+        _, dw_ref, _ = torch.ops.aten.convolution_backward(
+            g, x, w, [0], [stride, stride], [1, 1], [1, 1], False, [0, 0],
+            1, [False, True, False])
+        wgerr = (dw - dw_ref).abs().max().item() / max(
+            dw_ref.abs().max().item(), 1e-6)
         tag = "<<" if to < tm else ""
-        print(f"H{hgt:>3} C{c:>5} s{stride} {tm:9.3f} {to:9.3f}  "
-              f"err={err:.2e} {tag}")
+        print(f"H{hgt:>3} C{c:>5} s{stride} {tm:9.3f} {to:9.3f} "
+              f"{t_wg:8.3f} {t_wg_m:9.3f}  err={err:.2e} "
+              f"wgerr={wgerr:.2e} {tag}")
 
 
 if __name__ == "__main__":
