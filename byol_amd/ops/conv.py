@@ -253,18 +253,41 @@ class _Conv3x3Fn(torch.autograd.Function):
         # BYOL_C3WGRAD until the microbench proves per-shape wins
         use_wgrad = (os.environ.get("BYOL_C3WGRAD", "0") == "1"
                      and m % 32 == 0 and n % 64 == 0 and k % 64 == 0)
-        if use_wgrad:
+        # stride-1 dgrad == conv3x3(pad(dy), rot180(W)^T): reuses the fwd
+        # fast kernel directly — gated by BYOL_C3DGRAD
+        use_dgrad = (os.environ.get("BYOL_C3DGRAD", "0") == "1"
+                     and s == 1 and (b * hi * wi) % 128 == 0
+                     and k % 64 == 0 and n % 32 == 0)
+        if use_wgrad or use_dgrad:
             C = extension()
-            dx, _, _ = torch.ops.aten.convolution_backward(
-                dy, x, weight, [0], [s, s], [1, 1], [1, 1], False, [0, 0],
-                1, [True, False, False])
-            xpad = torch.empty(b * (hi + 2) * (wi + 2) * k,
-                               device=x.device, dtype=x.dtype)
-            C.pad_nhwc(_rows(x, k), xpad, b, hi, wi, k)
-            dw9 = torch.zeros(9 * n * k, device=x.device, dtype=x.dtype)
-            dw = torch.empty(n, k, 3, 3, device=x.device, dtype=x.dtype)
-            C.conv3x3_wgrad(_rows(dy, n), xpad, dw9, dw.reshape(-1), b, hi,
-                            wi, ho, wo, k, n, s)
+            if use_dgrad:
+                # Wp_d[tap][n][k] = W[n][k][2-ty][2-tx]
+                wpd = _cached_layout(
+                    weight, "3x3d",
+                    lambda: weight.flip([2, 3]).reshape(n, k, 9)
+                    .permute(2, 0, 1).contiguous())
+                dypad = torch.empty(b * (hi + 2) * (wi + 2) * n,
+                                    device=x.device, dtype=x.dtype)
+                C.pad_nhwc(_rows(dy, n), dypad, b, hi, wi, n)
+                dx = torch.empty_like(x)
+                C.conv3x3_fwd_fast(dypad, wpd, _rows(dx, k), b, hi, wi,
+                                   hi, wi, n, k, 1)
+            else:
+                dx, _, _ = torch.ops.aten.convolution_backward(
+                    dy, x, weight, [0], [s, s], [1, 1], [1, 1], False,
+                    [0, 0], 1, [True, False, False])
+            if use_wgrad:
+                xpad = torch.empty(b * (hi + 2) * (wi + 2) * k,
+                                   device=x.device, dtype=x.dtype)
+                C.pad_nhwc(_rows(x, k), xpad, b, hi, wi, k)
+                dw9 = torch.zeros(9 * n * k, device=x.device, dtype=x.dtype)
+                dw = torch.empty(n, k, 3, 3, device=x.device, dtype=x.dtype)
+                C.conv3x3_wgrad(_rows(dy, n), xpad, dw9, dw.reshape(-1), b,
+                                hi, wi, ho, wo, k, n, s)
+            else:
+                _, dw, _ = torch.ops.aten.convolution_backward(
+                    dy, x, weight, [0], [s, s], [1, 1], [1, 1], False,
+                    [0, 0], 1, [False, True, False])
             return dx, dw, None
         dx, dw, _ = torch.ops.aten.convolution_backward(
             dy, x, weight, [0], [s, s], [1, 1], [1, 1], False, [0, 0], 1,

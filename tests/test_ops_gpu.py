@@ -713,3 +713,29 @@ def test_conv3x3_wgrad_matches_oracle(ext, monkeypatch):
         assert torch.allclose(x.grad, x2.grad, rtol=1e-4, atol=1e-4)
         assert torch.allclose(w.grad, w2.grad, rtol=1e-3, atol=1e-2), \
             (hgt, c, stride, (w.grad - w2.grad).abs().max().item())
+
+
+def test_conv3x3_dgrad_via_fwd_matches_oracle(ext, monkeypatch):
+    """Stride-1 dgrad through the fwd fast kernel (rot180(W)^T on padded
+    dy) vs MIOpen."""
+    import torch.nn.functional as F
+    from byol_amd.ops.conv import _Conv3x3Fn
+    monkeypatch.setenv("BYOL_MFMA_CONV3X3", "1")
+    monkeypatch.setenv("BYOL_C3DGRAD", "1")
+    torch.manual_seed(35)
+    for bsz, hgt, c in [(8, 16, 128), (2, 32, 64)]:
+        if (bsz * hgt * hgt) % 128 != 0:
+            continue
+        x = torch.randn(bsz, c, hgt, hgt, device="cuda").to(
+            memory_format=torch.channels_last).requires_grad_(True)
+        w = (torch.randn(c, c, 3, 3, device="cuda") * 0.05
+             ).requires_grad_(True)
+        y = _Conv3x3Fn.apply(x, w, 1)
+        g = torch.randn_like(y).to(memory_format=torch.channels_last)
+        y.backward(g)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        F.conv2d(x2, w2, stride=1, padding=1).backward(g)
+        assert torch.allclose(x.grad, x2.grad, rtol=1e-4, atol=1e-4), \
+            (hgt, c, (x.grad - x2.grad).abs().max().item())
+        assert torch.allclose(w.grad, w2.grad, rtol=1e-3, atol=1e-2)

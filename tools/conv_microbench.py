@@ -155,6 +155,28 @@ def main_3x3(batch=512, iters=8):
         t_wg_m = timeit(lambda: torch.ops.aten.convolution_backward(
             g, x, w, [0], [stride, stride], [1, 1], [1, 1], False, [0, 0],
             1, [False, True, False]), iters)
+        # stride-1 dgrad via fwd kernel vs MIOpen dgrad-only
+        t_dgo = t_dgm = float("nan")
+        if stride == 1:
+            wpd = w.flip([2, 3]).reshape(c, c, 9).permute(2, 0, 1) \
+                .contiguous()
+            gpad = torch.empty(b * (hgt + 2) * (hgt + 2) * c,
+                               device="cuda")
+            dxo = torch.empty_like(x)
+            def dg_ours():
+                C.pad_nhwc(_rows(g, c), gpad, b, hgt, hgt, c)
+                C.conv3x3_fwd_fast(gpad, wpd, _rows(dxo, c), b, hgt, hgt,
+                                   hgt, hgt, c, c, 1)
+            t_dgo = timeit(dg_ours, iters)
+            t_dgm = timeit(lambda: torch.ops.aten.convolution_backward(
+                g, x, w, [0], [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+                [True, False, False]), iters)
+            dx_ref, _, _ = torch.ops.aten.convolution_backward(
+                g, x, w, [0], [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+                [True, False, False])
+            dg_ours()
+            dgerr = (dxo - dx_ref).abs().max().item()
+            assert dgerr < 1e-3, (hgt, c, dgerr)
         _, dw_ref, _ = torch.ops.aten.convolution_backward(
             g, x, w, [0], [stride, stride], [1, 1], [1, 1], False, [0, 0],
             1, [False, True, False])
@@ -162,8 +184,8 @@ def main_3x3(batch=512, iters=8):
             dw_ref.abs().max().item(), 1e-6)
         tag = "<<" if to < tm else ""
         print(f"H{hgt:>3} C{c:>5} s{stride} {tm:9.3f} {to:9.3f} "
-              f"{t_wg:8.3f} {t_wg_m:9.3f}  err={err:.2e} "
-              f"wgerr={wgerr:.2e} {tag}")
+              f"{t_wg:8.3f} {t_wg_m:9.3f} dg {t_dgo:7.3f}/{t_dgm:7.3f}  "
+              f"err={err:.2e} wgerr={wgerr:.2e} {tag}")
 
 
 if __name__ == "__main__":
