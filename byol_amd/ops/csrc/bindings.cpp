@@ -24,6 +24,17 @@ void launch_bn_bwd_reduce_v2(const float* dy, const float* y, const float* x,
                              const float* mean, const float* invstd,
                              float* red, int64_t m, int c, int relu,
                              int slot_mask, int grid, hipStream_t stream);
+void launch_bn_apply_v2(const float* x, const float* residual,
+                        const float* mean, const float* invstd,
+                        const float* weight, const float* bias, float* y,
+                        int64_t m, int c, int relu, int grid,
+                        hipStream_t stream);
+void launch_bn_bwd_apply_v2(const float* dy, const float* y, const float* x,
+                            const float* mean, const float* invstd,
+                            const float* weight, const float* red,
+                            float* dx, float* dresidual, float inv_count,
+                            int64_t m, int c, int relu, int grid,
+                            hipStream_t stream);
 void launch_bn_reduce_slots(const float* in, float* out, int n2c, int nslots,
                             hipStream_t stream);
 void launch_bn_finalize(const float* acc, float* mean, float* invstd,
@@ -254,6 +265,37 @@ void bn_bwd_reduce_v2(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
                           invstd.data_ptr<float>(), red.data_ptr<float>(),
                           m, (int)c, (int)relu, (int)(nslots - 1),
                           (int)grid, stream);
+}
+
+void bn_apply_v2(torch::Tensor x, c10::optional<torch::Tensor> residual,
+                 torch::Tensor mean, torch::Tensor invstd,
+                 torch::Tensor weight, torch::Tensor bias, torch::Tensor y,
+                 int64_t m, int64_t c, int64_t relu, int64_t grid) {
+  CHECK_IN(x); CHECK_IN(y);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_apply_v2(
+      x.data_ptr<float>(),
+      residual.has_value() ? residual->data_ptr<float>() : nullptr,
+      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+      weight.data_ptr<float>(), bias.data_ptr<float>(),
+      y.data_ptr<float>(), m, (int)c, (int)relu, (int)grid, stream);
+}
+
+void bn_bwd_apply_v2(torch::Tensor dy, torch::Tensor y, torch::Tensor x,
+                     torch::Tensor mean, torch::Tensor invstd,
+                     torch::Tensor weight, torch::Tensor red,
+                     torch::Tensor dx, c10::optional<torch::Tensor> dres,
+                     double inv_count, int64_t m, int64_t c, int64_t relu,
+                     int64_t grid) {
+  CHECK_IN(dy); CHECK_IN(dx);
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_bn_bwd_apply_v2(
+      dy.data_ptr<float>(), y.data_ptr<float>(), x.data_ptr<float>(),
+      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+      weight.data_ptr<float>(), red.data_ptr<float>(),
+      dx.data_ptr<float>(),
+      dres.has_value() ? dres->data_ptr<float>() : nullptr,
+      (float)inv_count, m, (int)c, (int)relu, (int)grid, stream);
 }
 
 void bn_reduce_slots(torch::Tensor in, torch::Tensor out, int64_t nslots) {
@@ -536,6 +578,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_stats", &bn_stats);
   mod.def("bn_stats_v2", &bn_stats_v2);
   mod.def("bn_bwd_reduce_v2", &bn_bwd_reduce_v2);
+  mod.def("bn_apply_v2", &bn_apply_v2);
+  mod.def("bn_bwd_apply_v2", &bn_bwd_apply_v2);
   mod.def("bn_reduce_slots", &bn_reduce_slots);
   mod.def("bn_finalize", &bn_finalize);
   mod.def("bn_apply", &bn_apply);

@@ -104,6 +104,50 @@ def main():
                 args.iters)
             if bestb[0] is None or t2 < bestb[0]:
                 bestb = (t2, grid)
+        # apply v1 vs v2 (fwd: read x + write y; bwd: 3 reads + 1 write)
+        weightp = torch.ones(c, device="cuda")
+        biasp = torch.zeros(c, device="cuda")
+        yout = torch.empty_like(x)
+        ta1 = bench(lambda: ext.bn_apply(x, None, mean, invstd, weightp,
+                                         biasp, yout, m, c, 1), args.iters)
+        besta = (None, None)
+        ext.bn_apply(x, None, mean, invstd, weightp, biasp, yout, m, c, 1)
+        ref_y = yout.clone()
+        for grid in (1024, 2048, 4096):
+            ext.bn_apply_v2(x, None, mean, invstd, weightp, biasp, yout,
+                            m, c, 1, grid)
+            assert torch.allclose(yout, ref_y, rtol=1e-5, atol=1e-5)
+            t2 = bench(lambda gd=grid: ext.bn_apply_v2(
+                x, None, mean, invstd, weightp, biasp, yout, m, c, 1, gd),
+                args.iters)
+            if besta[0] is None or t2 < besta[0]:
+                besta = (t2, grid)
+        red2c = torch.zeros(2 * c, device="cuda")
+        dxo = torch.empty_like(x)
+        tba1 = bench(lambda: ext.bn_bwd_apply(dy, y, x, mean, invstd,
+                                              weightp, red2c, dxo, None,
+                                              1.0 / m, m, c, 1), args.iters)
+        ext.bn_bwd_apply(dy, y, x, mean, invstd, weightp, red2c, dxo, None,
+                         1.0 / m, m, c, 1)
+        ref_dx = dxo.clone()
+        bestba = (None, None)
+        for grid in (1024, 2048, 4096):
+            ext.bn_bwd_apply_v2(dy, y, x, mean, invstd, weightp, red2c,
+                                dxo, None, 1.0 / m, m, c, 1, grid)
+            assert torch.allclose(dxo, ref_dx, rtol=1e-5, atol=1e-5)
+            t2 = bench(lambda gd=grid: ext.bn_bwd_apply_v2(
+                dy, y, x, mean, invstd, weightp, red2c, dxo, None, 1.0 / m,
+                m, c, 1, gd), args.iters)
+            if bestba[0] is None or t2 < bestba[0]:
+                bestba = (t2, grid)
+        apl_bytes = 2 * m * c * 4
+        bapl_bytes = 4 * m * c * 4
+        print(f"    apply {ta1 * 1e3:.3f}({apl_bytes / ta1 / 1e12:.2f}) -> "
+              f"{besta[0] * 1e3:.3f}({apl_bytes / besta[0] / 1e12:.2f}) "
+              f"g{besta[1]}   bwd_apply {tba1 * 1e3:.3f}"
+              f"({bapl_bytes / tba1 / 1e12:.2f}) -> {bestba[0] * 1e3:.3f}"
+              f"({bapl_bytes / bestba[0] / 1e12:.2f}) g{bestba[1]}")
+        del yout, dxo
         bwd_bytes = 3 * m * c * 4
         print(f"{m:>14} x {c:<5} {t1 * 1e3:>8.3f} "
               f"{bytes_read / t1 / 1e12:>8.2f} {best[0] * 1e3:>10.3f} "
