@@ -858,13 +858,22 @@ static bool wgrad_v3_enabled() {
   return f;
 }
 
+static int wgrad_blocks_cap() {
+  // sweepable block budget (BYOL_WGRAD_BLOCKS); 4096 measured default
+  static int cap = [] {
+    const char* v = getenv("BYOL_WGRAD_BLOCKS");
+    return v ? atoi(v) : 4096;
+  }();
+  return cap;
+}
+
 template <int TN, int TK>
 static void launch_wgrad_v3_tile(const float* dy, const float* x, float* dw,
                                  int64_t m, int n, int k,
                                  hipStream_t stream) {
   const int64_t tiles = (int64_t)(n / TN) * (k / TK);
   int64_t chunk = 32 * BK;
-  while (cdiv(m, chunk) * tiles > 4096) chunk *= 2;
+  while (cdiv(m, chunk) * tiles > wgrad_blocks_cap()) chunk *= 2;
   const int64_t grid = tiles * cdiv(m, chunk);
   if (conv_prio())
     hipLaunchKernelGGL((conv1x1_wgrad_v3_kernel<TN, TK, true>),
