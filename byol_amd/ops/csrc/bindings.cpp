@@ -106,6 +106,10 @@ void launch_pad_nhwc(const float* x, float* xp, int b, int hi, int wi,
 void launch_conv3x3_fwd_fast(const float* xp, const float* wp, float* y,
                              int b, int hi, int wi, int ho, int wo, int k,
                              int n, int stride, hipStream_t stream);
+void launch_conv3x3_fwd_nopad(const float* x, const float* wp, float* y,
+                              const float* zpage, int b, int hi, int wi,
+                              int ho, int wo, int k, int n, int stride,
+                              hipStream_t stream);
 void launch_conv3x3_wgrad(const float* dy, const float* xp, float* dw9,
                           float* dw, int b, int hi, int wi, int ho, int wo,
                           int k, int n, int stride, hipStream_t stream);
@@ -186,6 +190,19 @@ std::tuple<torch::Tensor, torch::Tensor> byol_loss_backward(
       grad_out.contiguous().data_ptr<float>(), g1.data_ptr<float>(),
       g2.data_ptr<float>(), n, batch, stream);
   return {g1, g2};
+}
+
+void conv3x3_fwd_nopad(torch::Tensor x, torch::Tensor wp, torch::Tensor y,
+                       torch::Tensor zpage, int64_t b, int64_t hi,
+                       int64_t wi, int64_t ho, int64_t wo, int64_t k,
+                       int64_t n, int64_t stride) {
+  CHECK_IN(x); CHECK_IN(wp); CHECK_IN(y); CHECK_IN(zpage);
+  TORCH_CHECK(zpage.numel() >= 4, "zpage must hold >= 16 bytes");
+  auto stream = at::hip::getCurrentHIPStream();
+  launch_conv3x3_fwd_nopad(x.data_ptr<float>(), wp.data_ptr<float>(),
+                           y.data_ptr<float>(), zpage.data_ptr<float>(),
+                           (int)b, (int)hi, (int)wi, (int)ho, (int)wo,
+                           (int)k, (int)n, (int)stride, stream);
 }
 
 void conv3x3_wgrad(torch::Tensor dy, torch::Tensor xp, torch::Tensor dw9,
@@ -564,6 +581,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("conv3x3_fwd", &conv3x3_fwd);
   mod.def("conv3x3_fwd_fast", &conv3x3_fwd_fast);
   mod.def("conv3x3_wgrad", &conv3x3_wgrad);
+  mod.def("conv3x3_fwd_nopad", &conv3x3_fwd_nopad);
   mod.def("pad_nhwc", &pad_nhwc);
   mod.def("conv1x1_fwd", &conv1x1_fwd);
   mod.def("conv1x1_dgrad", &conv1x1_dgrad);

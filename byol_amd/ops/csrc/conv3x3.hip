@@ -658,3 +658,169 @@ void launch_conv3x3_wgrad(const float* dy, const float* xp, float* dw9,
   hipLaunchKernelGGL(wgrad9_permute_kernel, dim3(grid_1d(nk, 256)),
                      dim3(256), 0, stream, dw9, dw, nk);
 }
+
+// ---------------------------------------------------------------------------
+// Pad-free fast forward: reads the UNPADDED input; out-of-bounds tap lanes
+// redirect their glds source to a 16-B zero page instead of a padded copy
+// (the pad pass cost 0.07-0.28 ms/call = most of the gap to MIOpen on the
+// stride-2 shapes).  Same pipeline as conv3x3_fwd_fast_kernel otherwise.
+// ---------------------------------------------------------------------------
+template <int TBN>
+__global__ __launch_bounds__(256)
+void conv3x3_fwd_nopad_kernel(const float* __restrict__ X,  // unpadded
+                              const float* __restrict__ Wp,  // [9][K][N]
+                              float* __restrict__ Y,
+                              const float* __restrict__ zpage,
+                              int B, int Hi, int Wi,
+                              int Ho, int Wo, int K, int N, int stride) {
+  constexpr int LDSB = 32 * TBN;
+  constexpr int LDSBUF = C3F_LDSW_A + LDSB;
+  __shared__ __attribute__((aligned(16))) float lds[2 * LDSBUF];
+
+  const int ntiles_n = N / TBN;
+  const int tile_m = blockIdx.x / ntiles_n;
+  const int tile_n = blockIdx.x % ntiles_n;
+  const int64_t m0 = (int64_t)tile_m * C3_BM;
+  const int n0 = tile_n * TBN;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  // wave tile: TBN=128 -> 2x2 quadrants of 64x64; TBN=64 -> 4x1 strips
+  const int wr = (TBN == 128) ? (wave >> 1) * 64 : wave * 32;
+  const int wc = (TBN == 128) ? (wave & 1) * 64 : 0;
+  constexpr int AT = (TBN == 128) ? 2 : 1;
+  constexpr int AU = 2;
+
+  const float* abase[4];
+  int aq[4], aiy[4], aix[4];
+  #pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int off = (wave * 4 + i) * 256 + lane * 4;
+    const int r = off >> 5;
+    const int q = (off & 31) >> 2;
+    aq[i] = ((q - r) & 7) << 2;
+    const int64_t gm = m0 + r;
+    const int64_t howo = (int64_t)Ho * Wo;
+    const int bb = (int)(gm / howo);
+    const int rem = (int)(gm % howo);
+    aiy[i] = (rem / Wo) * stride - 1;  // unpadded coords incl. pad offset
+    aix[i] = (rem % Wo) * stride - 1;
+    abase[i] = X + (((int64_t)bb * Hi + aiy[i]) * Wi + aix[i]) * K;
+  }
+
+  f32x16_c3 acc[AT][AU];
+  #pragma unroll
+  for (int t = 0; t < AT; ++t)
+    #pragma unroll
+    for (int u = 0; u < AU; ++u)
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) acc[t][u][e] = 0.f;
+
+  const int nk = K / C3_BK;
+  const int nsteps = 9 * nk;
+
+  auto stage = [&](int buf, int step) {
+    const int tap = step / nk;
+    const int k0 = (step % nk) * C3_BK;
+    const int dy = tap / 3, dx = tap % 3;
+    const int64_t tap_off = ((int64_t)dy * Wi + dx) * K;
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const bool ok = (unsigned)(aiy[i] + dy) < (unsigned)Hi &&
+                      (unsigned)(aix[i] + dx) < (unsigned)Wi;
+      const float* src =
+          ok ? abase[i] + tap_off + k0 + aq[i] : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * LDSBUF + (wave * 4 + i) * 256],
+          16, 0, 0);
+    }
+    const float* wtap = Wp + ((int64_t)tap * K + k0) * N;
+    #pragma unroll
+    for (int i = 0; i < TBN / 64; ++i) {
+      const int off = (wave * (TBN / 64) + i) * 256 + lane * 4;
+      const int r = off / TBN;
+      const int c = off % TBN;
+      const float* src = wtap + (int64_t)r * N + n0 + c;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) uint32_t*)src,
+          (__attribute__((address_space(3))) uint32_t*)
+              &lds[buf * LDSBUF + C3F_LDSW_A + (wave * (TBN / 64) + i)
+                   * 256],
+          16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int buf = 0;
+  for (int step = 0; step < nsteps; ++step) {
+    if (step + 1 < nsteps) stage(buf ^ 1, step + 1);
+    const float* la = &lds[buf * LDSBUF];
+    const float* lb = &lds[buf * LDSBUF + C3F_LDSW_A];
+    #pragma unroll
+    for (int kk = 0; kk < C3_BK; kk += 2) {
+      const int krow = kk + (lane >> 5);
+      #pragma unroll
+      for (int t = 0; t < AT; ++t) {
+        const int row = wr + t * 32 + (lane & 31);
+        const float a = la[row * 32 + ((((krow >> 2) + row) & 7) << 2)
+                           + (krow & 3)];
+        #pragma unroll
+        for (int u = 0; u < AU; ++u) {
+          const float b = lb[krow * TBN + wc + u * 32 + (lane & 31)];
+          acc[t][u] = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc[t][u],
+                                                           0, 0, 0);
+        }
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    buf ^= 1;
+  }
+
+  __syncthreads();
+  float* cw = &lds[wave * (TBN == 128 ? 4096 : 2048)];
+  #pragma unroll
+  for (int t = 0; t < AT; ++t)
+    #pragma unroll
+    for (int u = 0; u < AU; ++u)
+      #pragma unroll
+      for (int e = 0; e < 16; ++e) {
+        const int lrow = t * 32 + (e & 3) + 8 * (e >> 2) + 4 * (lane >> 5);
+        const int lcol = u * 32 + (lane & 31);
+        cw[lrow * 64 + lcol] = acc[t][u][e];
+      }
+  __builtin_amdgcn_s_barrier();
+  const int64_t gm_base = m0 + wr;
+  const int gn_base = n0 + wc;
+  #pragma unroll
+  for (int p = 0; p < (TBN == 128 ? 16 : 8); ++p) {
+    const int lrow = p * 4 + (lane >> 4);
+    const int lcol = (lane & 15) * 4;
+    const float4 v = *reinterpret_cast<const float4*>(&cw[lrow * 64 + lcol]);
+    *reinterpret_cast<float4*>(
+        &Y[(gm_base + lrow) * N + gn_base + lcol]) = v;
+  }
+}
+
+void launch_conv3x3_fwd_nopad(const float* x, const float* wp, float* y,
+                              const float* zpage, int b, int hi, int wi,
+                              int ho, int wo, int k, int n, int stride,
+                              hipStream_t stream) {
+  const int64_t m = (int64_t)b * ho * wo;
+  if (n % C3_BN == 0) {
+    const int64_t grid = (m / C3_BM) * (n / C3_BN);
+    hipLaunchKernelGGL((conv3x3_fwd_nopad_kernel<128>),
+                       dim3((uint32_t)grid), dim3(256), 0, stream, x, wp, y,
+                       zpage, b, hi, wi, ho, wo, k, n, stride);
+  } else {
+    const int64_t grid = (m / C3_BM) * (n / 64);
+    hipLaunchKernelGGL((conv3x3_fwd_nopad_kernel<64>),
+                       dim3((uint32_t)grid), dim3(256), 0, stream, x, wp, y,
+                       zpage, b, hi, wi, ho, wo, k, n, stride);
+  }
+}

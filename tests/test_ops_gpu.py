@@ -415,7 +415,7 @@ def test_conv3x3_fast_path_matches_oracle(ext, monkeypatch):
     monkeypatch.setenv("BYOL_CONV3X3_FAST", "1")
     torch.manual_seed(22)
     for bsz, hgt, c, stride in [(8, 16, 128, 1), (8, 16, 256, 2),
-                                (2, 32, 512, 1)]:
+                                (2, 32, 512, 1), (8, 16, 64, 1)]:
         x = torch.randn(bsz, c, hgt, hgt, device="cuda").to(
             memory_format=torch.channels_last)
         w = torch.randn(c, c, 3, 3, device="cuda") * 0.05
