@@ -738,15 +738,15 @@ void conv3x3_fwd_nopad_kernel(const float* __restrict__ X,  // unpadded
     }
     const float* wtap = Wp + ((int64_t)tap * K + k0) * N;
     #pragma unroll
-    for (int i = 0; i < TBN / 64; ++i) {
-      const int off = (wave * (TBN / 64) + i) * 256 + lane * 4;
+    for (int i = 0; i < TBN / 32; ++i) {  // 32*TBN floats / (4 waves * 1 KiB)
+      const int off = (wave * (TBN / 32) + i) * 256 + lane * 4;
       const int r = off / TBN;
       const int c = off % TBN;
       const float* src = wtap + (int64_t)r * N + n0 + c;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)src,
           (__attribute__((address_space(3))) uint32_t*)
-              &lds[buf * LDSBUF + C3F_LDSW_A + (wave * (TBN / 64) + i)
+              &lds[buf * LDSBUF + C3F_LDSW_A + (wave * (TBN / 32) + i)
                    * 256],
           16, 0, 0);
     }
