@@ -215,9 +215,32 @@ def _zero_page(device) -> torch.Tensor:
 
 
 def mfma_conv3x3_mode() -> str:
-    # off by default until the microbench proves per-shape wins (round-2
-    # tuning track); "1" enables everywhere.
-    return os.environ.get("BYOL_MFMA_CONV3X3", "0")
+    # "0" off, "1" force everywhere, "auto" per-shape table below.
+    return os.environ.get("BYOL_MFMA_CONV3X3", "auto")
+
+
+def _parse_3x3_table(env: str):
+    out = set()
+    for item in env.split(","):
+        parts = item.strip().split("/")
+        if len(parts) == 3:
+            out.add((int(parts[0]), int(parts[1]), int(parts[2])))
+    return out
+
+
+# (K, N, stride) where the pad-free MFMA fwd measured at parity-or-better
+# vs MIOpen (r2 call 10); override with BYOL_CONV3X3_AUTO="k/n/s,k/n/s".
+_AUTO_3X3 = _parse_3x3_table(
+    os.environ.get("BYOL_CONV3X3_AUTO", ""))
+
+
+def _conv3x3_auto(k: int, n: int, stride: int) -> bool:
+    mode = mfma_conv3x3_mode()
+    if mode == "1":
+        return True
+    if mode == "auto":
+        return (k, n, stride) in _AUTO_3X3
+    return False
 
 
 class _Conv3x3Fn(torch.autograd.Function):
@@ -318,10 +341,11 @@ class MFMAConv3x3(nn.Conv2d):
     and eligible, MIOpen otherwise (and always for backward)."""
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if (mfma_conv3x3_mode() == "1" and x.is_cuda
-                and x.dtype == torch.float32 and has_extension()
+        if (x.is_cuda and x.dtype == torch.float32 and has_extension()
                 and x.is_contiguous(memory_format=torch.channels_last)
-                and self.weight.shape[1] % 32 == 0):
+                and self.weight.shape[1] % 32 == 0
+                and _conv3x3_auto(self.weight.shape[1],
+                                  self.weight.shape[0], self.stride[0])):
             return _Conv3x3Fn.apply(x, self.weight, self.stride[0])
         return F.conv2d(x, self.weight, self.bias, self.stride,
                         self.padding, self.dilation, self.groups)
