@@ -7,7 +7,7 @@ import os
 
 # Ship the MI355X-tuned MIOpen find DB so first-run conv algo selection is
 # instant (same policy as bench.py; see byol_amd/ops/miopen_udb/).
-os.environ.setdefault("MIOPEN_FIND_MODE", "NORMAL")
+os.environ.setdefault("MIOPEN_FIND_MODE", "HYBRID")
 if "MIOPEN_USER_DB_PATH" not in os.environ:
     import shutil
     import tempfile
