@@ -39,7 +39,7 @@ def init_multiprocessing_and_cuda(rank: int, args):
     args.cuda = not args.no_cuda and torch.cuda.is_available()
     if args.cuda:
         # reference parity (cudnn.benchmark=True -> MIOpen exhaustive
-        # find); with MIOPEN_FIND_MODE set (main.py defaults it to NORMAL
+        # find); with MIOPEN_FIND_MODE set (main.py defaults it to HYBRID
         # + the shipped tuned DB) the fast find path is used instead
         torch.backends.cudnn.benchmark = (
             os.environ.get("MIOPEN_FIND_MODE", "") == "")
