@@ -119,3 +119,45 @@ def test_flat_ddp_broadcast_buffers_flag():
     running stats every forward."""
     import torch.multiprocessing as mp
     mp.spawn(_bcast_worker, nprocs=2, join=True)
+
+
+def test_conv3x3_auto_table_parsing():
+    from byol_amd.ops.conv import _parse_3x3_table
+    assert _parse_3x3_table("") == set()
+    assert _parse_3x3_table("128/128/2,64/64/1") == {(128, 128, 2),
+                                                     (64, 64, 1)}
+    assert _parse_3x3_table(" 256/256/1 ") == {(256, 256, 1)}
+
+
+def test_conv_dispatch_tables_consistent():
+    """Every wgrad-table shape is also dgrad-routed (the backward branch
+    assumes it), and tables only contain MFMA-eligible geometry."""
+    from byol_amd.ops.conv import _AUTO_DGRAD, _AUTO_SHAPES, _AUTO_WGRAD
+    for k, n in _AUTO_SHAPES | _AUTO_DGRAD | _AUTO_WGRAD:
+        assert k % 32 == 0 and n % 32 == 0, (k, n)
+    for k, n in _AUTO_WGRAD:
+        assert n % 64 == 0 and k % 64 == 0, (k, n)
+
+
+def test_half_dtype_fp16_cpu_plumbing(tmp_path):
+    """--half --half-dtype fp16 on CPU: scaler stays disabled, the engine
+    runs a debug step without error (the fp16 math itself is GPU-only)."""
+    import byol_amd.config as config
+    from byol_amd.engine import trainer
+
+    args = config.parse_args([
+        "--task", "synthetic_multi_augment_image_folder",
+        "--arch", "resnet18", "--representation-size", "512",
+        "--batch-size", "8", "--epochs", "1", "--num-replicas", "1",
+        "--no-cuda", "--half", "--half-dtype", "fp16", "--debug-step",
+        "--image-size-override", "32", "--synthetic-classes", "4",
+        "--workers-per-replica", "0",
+        "--model-dir", str(tmp_path / "m"),
+        "--log-dir", str(tmp_path / "l"),
+    ])
+    args.cuda = False
+    args.distributed_rank = 0
+    loader, model, grapher = trainer.build_loader_model_grapher(args)
+    from byol_amd.optim import build_optimizer
+    opt, sched = build_optimizer(model, args)
+    trainer.train(1, model, opt, loader.train_loader, grapher, args)
