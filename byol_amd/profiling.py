@@ -66,8 +66,12 @@ class PhaseTimer:
                 self.totals[phase] += s.elapsed_time(e) / 1000.0
             self._events.clear()
         total = sum(self.totals.values())
+        # known phases in canonical order, then any ad-hoc ones (e.g. the
+        # fused "step" phase of the round-2 engine loop)
+        keys = [k for k in self.PHASES if k in self.totals]
+        keys += [k for k in self.totals if k not in self.PHASES]
         parts = [f"{k}={self.totals.get(k, 0.0) * 1000:.1f}ms"
-                 for k in self.PHASES if k in self.totals]
+                 for k in keys]
         ips = self.samples / total if total > 0 else 0.0
         line = (f"[perf] {' '.join(parts)} total={total * 1000:.1f}ms "
                 f"images/sec={ips:.1f}")
