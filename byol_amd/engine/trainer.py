@@ -262,10 +262,17 @@ def execute_graph(epoch, model, loader, grapher, args, optimizer=None,
     GRAPH_WARM = 2
     graphed = None
 
+    batches = loader
+    if args.cuda:
+        # H2D on a dedicated copy stream, overlapped with compute
+        # (SURVEY.md K19); pass-through for device-resident batches
+        from .prefetch import DevicePrefetcher
+        batches = DevicePrefetcher(loader)
+
     timer.start("data")
     for num_minibatches, (augmentation1, augmentation2, labels) in \
-            enumerate(loader):
-        if args.cuda:
+            enumerate(batches):
+        if args.cuda and not augmentation1.is_cuda:
             augmentation1 = augmentation1.cuda(non_blocking=True)
             augmentation2 = augmentation2.cuda(non_blocking=True)
             labels = labels.cuda(non_blocking=True)

@@ -739,3 +739,34 @@ def test_conv3x3_dgrad_via_fwd_matches_oracle(ext, monkeypatch):
         assert torch.allclose(x.grad, x2.grad, rtol=1e-4, atol=1e-4), \
             (hgt, c, (x.grad - x2.grad).abs().max().item())
         assert torch.allclose(w.grad, w2.grad, rtol=1e-3, atol=1e-2)
+
+
+def test_engine_train_epoch_on_gpu(ext, tmp_path):
+    """The real trainer loop (setup -> lazy init -> train epoch -> test
+    epoch) on GPU with the synthetic task: exercises the device
+    prefetcher, fused BN/loss/CE/LARS through the engine, and the epoch
+    bookkeeping."""
+    import byol_amd.config as config
+    from byol_amd.engine import trainer
+    from byol_amd.optim import build_optimizer
+
+    args = config.parse_args([
+        "--task", "synthetic_multi_augment_image_folder",
+        "--arch", "resnet18", "--representation-size", "512",
+        "--batch-size", "16", "--epochs", "1", "--num-replicas", "1",
+        "--image-size-override", "32", "--synthetic-classes", "8",
+        "--synthetic-train-samples", "64", "--synthetic-test-samples", "32",
+        "--workers-per-replica", "0", "--channels-last",
+        "--model-dir", str(tmp_path / "m"),
+        "--log-dir", str(tmp_path / "l"),
+    ])
+    args.cuda = True
+    args.distributed_rank = 0
+    loader, model, grapher = trainer.build_loader_model_grapher(args)
+    opt, sched = build_optimizer(model, args)
+    trainer.train(1, model, opt, loader.train_loader, grapher, args)
+    loss = trainer.test(1, model, loader.test_loader, grapher, args)
+    assert loss == loss  # finite, not NaN
+    flat = (model.module if hasattr(model, "module") else model) \
+        .flat_space.flat_params
+    assert torch.isfinite(flat).all()
