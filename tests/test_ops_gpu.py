@@ -770,3 +770,29 @@ def test_engine_train_epoch_on_gpu(ext, tmp_path):
     flat = (model.module if hasattr(model, "module") else model) \
         .flat_space.flat_params
     assert torch.isfinite(flat).all()
+
+
+def test_engine_half_bf16_epoch_on_gpu(ext, tmp_path):
+    """--half (bf16 autocast + bf16-I/O fused BN) through the real engine."""
+    import byol_amd.config as config
+    from byol_amd.engine import trainer
+    from byol_amd.optim import build_optimizer
+
+    args = config.parse_args([
+        "--task", "synthetic_multi_augment_image_folder",
+        "--arch", "resnet18", "--representation-size", "512",
+        "--batch-size", "16", "--epochs", "1", "--num-replicas", "1",
+        "--image-size-override", "32", "--synthetic-classes", "8",
+        "--synthetic-train-samples", "48", "--synthetic-test-samples", "16",
+        "--workers-per-replica", "0", "--channels-last", "--half",
+        "--model-dir", str(tmp_path / "m"),
+        "--log-dir", str(tmp_path / "l"),
+    ])
+    args.cuda = True
+    args.distributed_rank = 0
+    loader, model, grapher = trainer.build_loader_model_grapher(args)
+    opt, sched = build_optimizer(model, args)
+    trainer.train(1, model, opt, loader.train_loader, grapher, args)
+    flat = (model.module if hasattr(model, "module") else model) \
+        .flat_space.flat_params
+    assert torch.isfinite(flat).all()
