@@ -107,6 +107,23 @@ class FlatParamSpace:
             out[name[len(prefix):]] = buffer[offset:offset + n].view(shape)
         return out
 
+    def verify(self) -> None:
+        """Raise if any parameter is no longer a view of the flat buffer —
+        catches the silent footgun of a device/memory-format move AFTER
+        ``finalize()`` (``Module.to`` rebinds ``param.data`` to fresh
+        storage, detaching it from the buffer the fused EMA/LARS/DDP ops
+        update)."""
+        base_ptr = self.flat_params.data_ptr()
+        end_ptr = base_ptr + self.flat_params.numel() \
+            * self.flat_params.element_size()
+        for name, p in self.module.named_parameters():
+            ptr = p.data_ptr()
+            if not (base_ptr <= ptr < end_ptr):
+                raise RuntimeError(
+                    f"parameter {name!r} is detached from the flat buffer "
+                    "— structural changes (.to, SyncBN conversion, memory-"
+                    "format moves) must happen BEFORE finalize()")
+
     def zero_grads(self) -> None:
         self.flat_grads.zero_()
 

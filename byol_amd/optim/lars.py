@@ -77,7 +77,11 @@ class LARS(Optimizer):
         self.optim.add_param_group(param_group)
 
     def attach_flat_space(self, flat_space):
-        """Enable the fused flat step and flat zero_grad."""
+        """Enable the fused flat step and flat zero_grad.  Verifies every
+        parameter still views the flat buffer (catches post-``finalize``
+        device/format moves early, instead of silently training a model
+        whose weights the optimizer no longer touches)."""
+        flat_space.verify()
         self._flat_space = flat_space
 
     @torch.no_grad()

@@ -156,3 +156,22 @@ def test_state_dict_roundtrip_preserves_flat_space():
     p = m2.linear_classifier.weight
     assert p.data_ptr() >= m2.flat_space.flat_params.data_ptr()
     assert m2.target_network.step == m.target_network.step
+
+
+def test_flat_space_verify_catches_detached_params():
+    """Module.to(memory_format=...) AFTER finalize() rebinds param.data off
+    the flat buffer; verify() must catch it (the fused EMA/LARS/DDP ops
+    would otherwise silently update a buffer the model no longer reads)."""
+    import pytest
+    import torch
+
+    from byol_amd.models.byol import BYOL
+
+    m = BYOL(arch="resnet18", base_network_output_size=512,
+             projection_output_size=32, classifier_output_size=4,
+             total_training_steps=10, head_latent_size=64)
+    m.finalize()
+    m.flat_space.verify()  # intact after finalize
+    m.base_network.to(memory_format=torch.channels_last)  # the footgun
+    with pytest.raises(RuntimeError, match="detached"):
+        m.flat_space.verify()
