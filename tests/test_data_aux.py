@@ -181,3 +181,36 @@ def test_mnist_task(tmp_path):
     assert loader.output_size == 10
     a1, a2, labels = next(iter(loader.train_loader))
     assert a1.shape == (4, 3, 16, 16)  # mnist expanded to 3 channels
+
+
+def test_saver_restore_reseeds_best(tmp_path):
+    """After resume, the pre-restart best loss must survive: a worse
+    post-restart epoch may NOT overwrite the best checkpoint."""
+    import types
+
+    import torch.nn as nn
+
+    from byol_amd.engine.saver import CheckpointBundle, ModelSaver
+
+    args = types.SimpleNamespace(model_dir=str(tmp_path), uid="t",
+                                 arch="a", batch_size=1, num_replicas=1,
+                                 optimizer="sgd")
+    model = nn.Linear(2, 2)
+    bundle = CheckpointBundle(model, None, None, args)
+    saver = ModelSaver(bundle, early_stop=False, rank=0, burn_in_interval=0)
+    assert saver(1.0) is False  # epoch 1: best=1.0, checkpoint written
+    assert saver(2.0) is False  # worse: stale
+
+    # simulate restart
+    saver2 = ModelSaver(bundle, early_stop=False, rank=0,
+                        burn_in_interval=0)
+    state = saver2.restore()
+    assert state["best_loss"] == 1.0
+    assert saver2.best == 1.0
+    # a worse epoch after resume must NOT count as improved
+    before = bundle.path()
+    import os
+    mtime = os.path.getmtime(before)
+    assert saver2(1.5) is False
+    assert saver2.best == 1.0
+    assert os.path.getmtime(before) == mtime  # best checkpoint untouched
