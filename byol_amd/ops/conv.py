@@ -66,7 +66,7 @@ def _eligible(x: torch.Tensor, weight: torch.Tensor) -> bool:
     mode = mfma_conv_mode()
     if mode == "0":
         return False
-    if mode == "1":
+    if mode in ("1", "autotune"):
         return True
     shape = (k, weight.shape[0])
     return shape in _AUTO_SHAPES or shape in _AUTO_DGRAD
@@ -120,7 +120,9 @@ class _Conv1x1Fn(torch.autograd.Function):
             weight, "1x1",
             lambda: (lambda v: (v, v.t().contiguous()))(
                 weight.reshape(n, k).contiguous()))
-        use_ours = mfma_conv_mode() == "1" or (k, n) in _AUTO_SHAPES
+        mode = mfma_conv_mode()
+        use_ours = (mode == "1" or (k, n) in _AUTO_SHAPES
+                    or (mode == "autotune"))
         if use_ours:
             y = torch.empty((b, n, h, w), device=x.device, dtype=x.dtype,
                             memory_format=torch.channels_last)
@@ -186,12 +188,45 @@ class _Conv1x1Fn(torch.autograd.Function):
         return dx, dw
 
 
+@torch.no_grad()
+def _autotune_decide_1x1(x: torch.Tensor, weight: torch.Tensor) -> bool:
+    """cudnn.benchmark-style runtime selection (BYOL_MFMA_CONV1X1=autotune):
+    first sight of a (k, n, m) times our fwd kernel vs MIOpen, cached for
+    the process (and BYOL_AUTOTUNE_CACHE if set)."""
+    from .autotune import autotuner
+    b, k, h, w = x.shape
+    n = weight.shape[0]
+    m = b * h * w
+    tun = autotuner()
+    key = ("c1f", k, n, m)
+    hit = tun.decisions.get(key)
+    if hit is not None:
+        return hit
+    C = extension()
+    xd = x.detach()
+    wd = weight.detach()
+    xr = _rows(xd, k)
+    wv = wd.reshape(n, k).contiguous()
+    wt = wv.t().contiguous()
+    yr = torch.empty(m, n, device=x.device, dtype=x.dtype)
+    return tun.choose(
+        key,
+        ours=lambda: C.conv1x1_fwd(xr, wv, wt, yr, m, k, n),
+        theirs=lambda: F.conv2d(xd, wd))
+
+
 class MFMAConv1x1(nn.Conv2d):
     """Drop-in for stride-1 bias-free 1x1 nn.Conv2d; routes to the MFMA
-    kernels when eligible, MIOpen otherwise."""
+    kernels when eligible, MIOpen otherwise.  Dispatch modes
+    (BYOL_MFMA_CONV1X1): "auto" = measured static tables (default),
+    "autotune" = runtime per-shape selection, "1" force, "0" off."""
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if _eligible(x, self.weight):
+            if (mfma_conv_mode() == "autotune"
+                    and not _autotune_decide_1x1(x, self.weight)):
+                return F.conv2d(x, self.weight, self.bias, self.stride,
+                                self.padding, self.dilation, self.groups)
             return _Conv1x1Fn.apply(x, self.weight)
         return F.conv2d(x, self.weight, self.bias, self.stride,
                         self.padding, self.dilation, self.groups)
@@ -336,16 +371,40 @@ class _Conv3x3Fn(torch.autograd.Function):
         return dx, dw, None
 
 
+@torch.no_grad()
+def _autotune_decide_3x3(x: torch.Tensor, weight: torch.Tensor,
+                         stride: int) -> bool:
+    from .autotune import autotuner
+    b, k, h, w = x.shape
+    n = weight.shape[0]
+    tun = autotuner()
+    key = ("c3f", k, n, stride, b * h * w)
+    hit = tun.decisions.get(key)
+    if hit is not None:
+        return hit
+    xd = x.detach()
+    wd = weight.detach()
+    return tun.choose(
+        key,
+        ours=lambda: _Conv3x3Fn.apply(xd, wd, stride),
+        theirs=lambda: F.conv2d(xd, wd, stride=stride, padding=1))
+
+
 class MFMAConv3x3(nn.Conv2d):
     """Drop-in for bias-free pad-1 3x3 nn.Conv2d; MFMA forward when enabled
-    and eligible, MIOpen otherwise (and always for backward)."""
+    and eligible, MIOpen otherwise (and always for backward unless the
+    BYOL_C3WGRAD/BYOL_C3DGRAD gates are set)."""
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if (x.is_cuda and x.dtype == torch.float32 and has_extension()
                 and x.is_contiguous(memory_format=torch.channels_last)
-                and self.weight.shape[1] % 32 == 0
-                and _conv3x3_auto(self.weight.shape[1],
-                                  self.weight.shape[0], self.stride[0])):
-            return _Conv3x3Fn.apply(x, self.weight, self.stride[0])
+                and self.weight.shape[1] % 32 == 0):
+            mode = mfma_conv3x3_mode()
+            take = _conv3x3_auto(self.weight.shape[1],
+                                 self.weight.shape[0], self.stride[0])
+            if not take and mode == "autotune":
+                take = _autotune_decide_3x3(x, self.weight, self.stride[0])
+            if take:
+                return _Conv3x3Fn.apply(x, self.weight, self.stride[0])
         return F.conv2d(x, self.weight, self.bias, self.stride,
                         self.padding, self.dilation, self.groups)

@@ -161,3 +161,37 @@ def test_half_dtype_fp16_cpu_plumbing(tmp_path):
     from byol_amd.optim import build_optimizer
     opt, sched = build_optimizer(model, args)
     trainer.train(1, model, opt, loader.train_loader, grapher, args)
+
+
+def test_autotuner_caches_and_persists(tmp_path):
+    """Decision caching + JSON persistence with an injected fake timer."""
+    from byol_amd.ops.autotune import Autotuner
+
+    calls = []
+
+    def fake_timer(fn):
+        calls.append(fn)
+        fn()
+        return 1.0 if getattr(fn, "_fast", False) else 2.0
+
+    path = str(tmp_path / "tune.json")
+    tun = Autotuner(cache_path=path, timer=fake_timer)
+
+    def ours():
+        pass
+    ours._fast = True
+
+    def theirs():
+        pass
+
+    assert tun.choose(("c1f", 64, 64, 1024), ours, theirs) is True
+    assert len(calls) == 2
+    # cached: no more timing
+    assert tun.choose(("c1f", 64, 64, 1024), ours, theirs) is True
+    assert len(calls) == 2
+
+    # persisted decisions reload
+    tun2 = Autotuner(cache_path=path, timer=fake_timer)
+    assert tun2.decisions[("c1f", 64, 64, 1024)] is True
+    assert tun2.choose(("c1f", 64, 64, 1024), theirs, ours) is True
+    assert len(calls) == 2
