@@ -9,6 +9,7 @@ the compute stream.  Pass-through when batches are already on device (the
 GPU-augmentation iterable) or when CUDA is unavailable.
 """
 
+from collections import deque
 from typing import Iterable, Iterator, Tuple
 
 import torch
@@ -21,6 +22,10 @@ class DevicePrefetcher:
         self.loader = loader
         self.device = device or torch.device("cuda")
         self.stream = torch.cuda.Stream(device=self.device)
+        # pinned host batches stay referenced for two iterations so an
+        # in-flight async H2D never reads freed host memory (belt and
+        # braces on top of torch's host-allocator event tracking)
+        self._keepalive = deque(maxlen=2)
 
     def __len__(self):
         return len(self.loader)
@@ -45,6 +50,7 @@ class DevicePrefetcher:
                 yield batch
                 continue
             moved = self._upload(batch)
+            self._keepalive.append(batch)
             ev = torch.cuda.Event()
             ev.record(self.stream)
             if pending is not None:
