@@ -134,3 +134,53 @@ def test_sync_bn_matches_full_batch_bn(tmp_path):
                           atol=1e-4)
     assert torch.allclose(r0["rm"], bn.running_mean, atol=1e-5)
     assert torch.allclose(r0["rv"], bn.running_var, atol=1e-4)
+
+
+def _stress4_worker(rank, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=4)
+    from byol_amd import layers
+    from byol_amd.optim.lars import LARS
+    from byol_amd.parallel.ddp import FlatDDP
+
+    model = _small_byol(seed=rank)
+    model.finalize()
+    ddp = FlatDDP(model, bucket_cap_mb=0.02)  # many tiny buckets
+    inner = torch.optim.SGD(layers.add_weight_decay(model, 1e-6), lr=0.05,
+                            momentum=0.9)
+    opt = LARS(inner, eps=0.0)
+    opt.attach_flat_space(model.flat_space)
+    ddp.train()
+    torch.manual_seed(7)
+    x1 = torch.rand(8, 3, 32, 32)
+    x2 = torch.rand(8, 3, 32, 32)
+    for step in range(3):
+        a1 = x1[rank * 2:(rank + 1) * 2]
+        a2 = x2[rank * 2:(rank + 1) * 2]
+        out = ddp(a1, a2)
+        loss = out["online_prediction1"].square().mean() + \
+            out["linear_preds"].square().mean()
+        opt.zero_grad()
+        loss.backward()
+        ddp.finish_grad_sync()
+        opt.step()
+    torch.save(model.flat_space.flat_params.clone(),
+               os.path.join(out_dir, f"p{rank}.pt"))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_flat_ddp_four_rank_stress(tmp_path):
+    """4 ranks x 3 full steps (forward+backward+LARS) over many tiny
+    buckets: parameters must stay bit-identical across replicas (the
+    replicated-state invariant the 8-GPU run depends on)."""
+    port = 29517
+    mp.spawn(_stress4_worker, args=(port, str(tmp_path)), nprocs=4,
+             join=True)
+    ps = [torch.load(os.path.join(str(tmp_path), f"p{r}.pt"),
+                     weights_only=True) for r in range(4)]
+    for r in range(1, 4):
+        assert torch.equal(ps[0], ps[r]), \
+            f"rank {r} diverged: max|d|=" \
+            f"{(ps[0] - ps[r]).abs().max().item()}"
