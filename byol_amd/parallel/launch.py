@@ -36,6 +36,16 @@ def init_multiprocessing_and_cuda(rank: int, args):
         os.environ["CUDA_VISIBLE_DEVICES"] = str(rank)
         args.distributed_rank = rank
 
+    # per-rank MIOpen user-DB copy: ranks sharing one sqlite DB contend on
+    # its lock (main.py sets a single parent dir before spawn)
+    udb = os.environ.get("MIOPEN_USER_DB_PATH")
+    if udb and args.num_replicas > 1 and not udb.endswith(f".r{rank}"):
+        import shutil
+        rank_dir = f"{udb}.r{rank}"
+        if os.path.isdir(udb):
+            shutil.copytree(udb, rank_dir, dirs_exist_ok=True)
+            os.environ["MIOPEN_USER_DB_PATH"] = rank_dir
+
     args.cuda = not args.no_cuda and torch.cuda.is_available()
     if args.cuda:
         # reference parity (cudnn.benchmark=True -> MIOpen exhaustive
