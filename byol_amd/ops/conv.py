@@ -148,7 +148,7 @@ class _Conv1x1Fn(torch.autograd.Function):
             dx = torch.empty_like(x)
             C.conv1x1_dgrad(_rows(dy, n), wv, _rows(dx, k), m, n, k)
             if os.environ.get("BYOL_WGRAD", "atomic") == "v2":
-                # round-2 candidate: partial slabs + reduce (no atomics)
+                # v2 (measured slower; kept for A/Bs): partial slabs + reduce
                 nchunks = C.wgrad_nchunks(m, n, k)
                 partial = torch.empty(nchunks * n * k, device=x.device,
                                       dtype=x.dtype)

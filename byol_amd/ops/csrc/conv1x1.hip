@@ -519,7 +519,8 @@ void conv1x1_gemm_fast64_kernel(const float* __restrict__ A,
 }
 
 // ---------------------------------------------------------------------------
-// wgrad v2 (round-2 candidate): per-chunk PARTIAL slabs with plain stores +
+// wgrad v2 (measured r2: SLOWER than the v1 atomics on every shape —
+// kept for the record; v3 below is the default): PARTIAL slabs with plain stores +
 // a reduce kernel, replacing the atomic epilogue (64 atomic RMWs per lane
 // were the gap vs MIOpen).  partial layout: [chunk][N][K] fp32.
 // ---------------------------------------------------------------------------

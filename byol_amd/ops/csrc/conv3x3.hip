@@ -142,7 +142,8 @@ void conv3x3_fwd_kernel(const float* __restrict__ X,
 }
 
 // ---------------------------------------------------------------------------
-// Fast path (round-2 candidate): zero-padded input copy makes every tap an
+// Fast path (validated r2; superseded by the pad-free variant below
+// for dispatch): zero-padded input copy makes every tap an
 // unguarded uniform shift, so A and B both stage through glds
 // (global_load_lds) with a 2-deep double buffer across the (tap, k0)
 // sequence — the conv1x1 fast-path structure with 9x the K depth.
