@@ -138,6 +138,15 @@ def build_parser() -> argparse.ArgumentParser:
                         help="NHWC memory format for conv throughput")
     parser.add_argument("--perf-stats", action="store_true", default=False,
                         help="per-phase HIP-event timing summary per epoch")
+    parser.add_argument("--conv-dispatch", type=str, default=None,
+                        choices=["auto", "autotune", "force", "off"],
+                        help="MFMA conv dispatch: auto = measured per-shape "
+                             "tables (default), autotune = runtime "
+                             "ours-vs-MIOpen selection per new shape "
+                             "(cudnn.benchmark-style; cache via "
+                             "BYOL_AUTOTUNE_CACHE), force/off = always/"
+                             "never our kernels.  Sets BYOL_MFMA_CONV1X1/"
+                             "CONV3X3 for this process")
     parser.add_argument("--hip-graph", action="store_true", default=False,
                         help="capture the training step as one hipGraph and "
                              "replay it (single-GPU, no --polyak-ema; "
@@ -152,4 +161,11 @@ def parse_args(argv=None):
         del args.__dict__["synthetic_train_samples"]
     if getattr(args, "synthetic_test_samples", 1) is None:
         del args.__dict__["synthetic_test_samples"]
+    if args.conv_dispatch is not None:
+        import os
+        mode = {"auto": "auto", "autotune": "autotune", "force": "1",
+                "off": "0"}[args.conv_dispatch]
+        os.environ["BYOL_MFMA_CONV1X1"] = mode
+        # 3x3 default is the (empty) auto table; force/off/autotune map 1:1
+        os.environ["BYOL_MFMA_CONV3X3"] = mode
     return args

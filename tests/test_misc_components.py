@@ -195,3 +195,17 @@ def test_autotuner_caches_and_persists(tmp_path):
     assert tun2.decisions[("c1f", 64, 64, 1024)] is True
     assert tun2.choose(("c1f", 64, 64, 1024), theirs, ours) is True
     assert len(calls) == 2
+
+
+def test_conv_dispatch_flag_sets_env(monkeypatch):
+    import os
+
+    import byol_amd.config as config
+
+    monkeypatch.delenv("BYOL_MFMA_CONV1X1", raising=False)
+    monkeypatch.delenv("BYOL_MFMA_CONV3X3", raising=False)
+    config.parse_args(["--conv-dispatch", "autotune"])
+    assert os.environ["BYOL_MFMA_CONV1X1"] == "autotune"
+    assert os.environ["BYOL_MFMA_CONV3X3"] == "autotune"
+    config.parse_args(["--conv-dispatch", "force"])
+    assert os.environ["BYOL_MFMA_CONV1X1"] == "1"
