@@ -322,7 +322,7 @@ __global__ void bn_bwd_apply_bf16_kernel(const ushort* __restrict__ dy,
 }
 
 // ---------------------------------------------------------------------------
-static int stats_grid_bf(int64_t m, int c) {
+__attribute__((unused)) static int stats_grid_bf(int64_t m, int c) {
   const int c4 = c >> 2;
   const int cw = c4 < 256 ? c4 : 256;
   const int rpb = 256 / cw;

@@ -33,7 +33,7 @@ __global__ void ce_topk_fwd_kernel(const float* __restrict__ logits,
   __syncthreads();
   if (threadIdx.x == 0) {
     float v = sred[0];
-    for (int w = 1; w < blockDim.x / WAVE_SIZE; ++w) v = fmaxf(v, sred[w]);
+    for (int w = 1; w < (int)(blockDim.x / WAVE_SIZE); ++w) v = fmaxf(v, sred[w]);
     sred[0] = v;
   }
   __syncthreads();
@@ -72,7 +72,8 @@ __global__ void ce_topk_fwd_kernel(const float* __restrict__ logits,
 
   if (threadIdx.x == 0) {
     float total = 0.f;
-    for (int w = 0; w < blockDim.x / WAVE_SIZE; ++w) total += ssum[w];
+    for (int w = 0; w < (int)(blockDim.x / WAVE_SIZE); ++w)
+      total += ssum[w];
     const float lse = logf(total) + vmax;
     const float lv = slabel;
     const float loss = lse - lv;
@@ -84,7 +85,7 @@ __global__ void ce_topk_fwd_kernel(const float* __restrict__ logits,
     // thread's local top-5 unless that thread saw >5 greater values; for
     // rank<=5 checks the pool is sufficient).
     int greater = 0;
-    for (int t = 0; t < blockDim.x * 5; ++t)
+    for (int t = 0; t < (int)(blockDim.x * 5); ++t)
       if (stop[t] > lv) ++greater;
     if (greater > 5) greater = 5;  // only need <1 and <5 decisions
     atomicAdd(&out[0], loss);

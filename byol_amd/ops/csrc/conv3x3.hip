@@ -185,9 +185,9 @@ void conv3x3_fwd_fast_kernel(const float* __restrict__ Xp,  // padded NHWC
                              float* __restrict__ Y,
                              int B, int Hp, int Wpp,  // padded dims
                              int Ho, int Wo, int K, int N, int stride) {
+  (void)B;  // batch rides inside the flattened m decomposition
   __shared__ __attribute__((aligned(16))) float lds[2 * C3F_LDSW_BUF];
 
-  const int64_t M = (int64_t)B * Ho * Wo;
   const int ntiles_n = N / C3_BN;
   const int bid = XSWZ ? xcd_remap(blockIdx.x, gridDim.x) : blockIdx.x;
   const int tile_m = bid / ntiles_n;
@@ -331,9 +331,9 @@ void conv3x3_fwd_fast64_kernel(const float* __restrict__ Xp,
                                float* __restrict__ Y,
                                int B, int Hp, int Wpp,
                                int Ho, int Wo, int K, int N, int stride) {
+  (void)B;  // batch rides inside the flattened m decomposition
   __shared__ __attribute__((aligned(16))) float lds[2 * C3F64_LDSW_BUF];
 
-  const int64_t M = (int64_t)B * Ho * Wo;
   const int ntiles_n = N / 64;
   const int bid = XSWZ ? xcd_remap(blockIdx.x, gridDim.x) : blockIdx.x;
   const int tile_m = bid / ntiles_n;
@@ -674,6 +674,7 @@ void conv3x3_fwd_nopad_kernel(const float* __restrict__ X,  // unpadded
                               const float* __restrict__ zpage,
                               int B, int Hi, int Wi,
                               int Ho, int Wo, int K, int N, int stride) {
+  (void)B;  // batch rides inside the flattened m decomposition
   constexpr int LDSB = 32 * TBN;
   constexpr int LDSBUF = C3F_LDSW_A + LDSB;
   __shared__ __attribute__((aligned(16))) float lds[2 * LDSBUF];

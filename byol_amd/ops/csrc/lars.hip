@@ -105,7 +105,7 @@ void launch_lars_momentum_step(float* p, const float* g, float* m,
                                const float* lr_dev,
                                float momentum, int m_init,
                                hipStream_t stream) {
-  hipMemsetAsync(norm_acc, 0, sizeof(float) * 2 * nseg, stream);
+  (void)hipMemsetAsync(norm_acc, 0, sizeof(float) * 2 * nseg, stream);
   hipLaunchKernelGGL(lars_norms_kernel, dim3(nchunks), dim3(256), 0, stream,
                      p, g, norm_acc, seg_off, seg_len, seg_wd, chunk_seg,
                      chunk_base, chunk, nchunks);
