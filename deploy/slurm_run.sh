@@ -15,7 +15,7 @@ MASTER=$(squeue -j "${SLURM_ARRAY_JOB_ID}" -O minnodelist,arraytaskid -h |
 RANK=${SLURM_ARRAY_TASK_ID}
 
 export HSA_ENABLE_IPC_MODE_LEGACY=0
-export MIOPEN_FIND_MODE=${MIOPEN_FIND_MODE:-NORMAL}
+export MIOPEN_FIND_MODE=${MIOPEN_FIND_MODE:-HYBRID}
 
 exec python main.py \
   --task multi_augment_image_folder \
