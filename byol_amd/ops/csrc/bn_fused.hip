@@ -554,7 +554,6 @@ __global__ void bn_stats_v2_kernel(const float* __restrict__ x,
     }
     // after the in-wave fold, row-lane tr % rows_in_wave == 0 holds the
     // wave-partial; cross-wave partials go through LDS
-    const bool wave_leader = active && (tr % rows_in_wave) == 0;
     scratch[threadIdx.x].s = ts;
     scratch[threadIdx.x].ss = tss;
     __syncthreads();
@@ -580,7 +579,6 @@ __global__ void bn_stats_v2_kernel(const float* __restrict__ x,
       }
     }
     __syncthreads();
-    (void)wave_leader;
   }
 }
 
