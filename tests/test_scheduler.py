@@ -89,3 +89,29 @@ def test_scheduler_state_roundtrip():
     assert sched2.get_last_lr() == pytest.approx(sched.get_last_lr())
     assert opt2.param_groups[0]["lr"] == pytest.approx(
         opt.param_groups[0]["lr"])
+
+
+def test_scheduler_resume_mid_warmup():
+    """Resume inside the warmup phase: the rewritten LinearWarmup must
+    restore its step counter and keep ramping from where it stopped."""
+    model = nn.Linear(4, 4)
+    args = make_args(warmup=4, epochs=12)
+    opt = torch.optim.SGD(model.parameters(), lr=1.0)
+    sched = build_lr_schedule(opt, args)
+    sched.step()  # -> 1/4
+    sd = sched.state_dict()
+    opt_sd = opt.state_dict()
+
+    opt2 = torch.optim.SGD(model.parameters(), lr=1.0)
+    sched2 = build_lr_schedule(opt2, args)
+    opt2.load_state_dict(opt_sd)
+    sched2.load_state_dict(sd)
+    assert sched2.warmup.complete is False
+    assert sched2.warmup.last_epoch == 1
+    sched2.step()
+    assert opt2.param_groups[0]["lr"] == pytest.approx(2.0 / 4)
+    sched2.step()
+    assert opt2.param_groups[0]["lr"] == pytest.approx(3.0 / 4)
+    sched2.step()
+    assert opt2.param_groups[0]["lr"] == pytest.approx(1.0)
+    assert sched2.warmup.complete is True
