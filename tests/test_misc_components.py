@@ -209,3 +209,25 @@ def test_conv_dispatch_flag_sets_env(monkeypatch):
     assert os.environ["BYOL_MFMA_CONV3X3"] == "autotune"
     config.parse_args(["--conv-dispatch", "force"])
     assert os.environ["BYOL_MFMA_CONV1X1"] == "1"
+
+
+def test_topk_metric_cpu_oracle():
+    """topk vs a brute-force ranking check, incl. k > num_classes clamp."""
+    import torch
+
+    from byol_amd.engine import metrics
+
+    torch.manual_seed(3)
+    logits = torch.randn(64, 10)
+    labels = torch.randint(10, (64,))
+    t1, t5 = metrics.topk(logits, labels, topk=(1, 5))
+    order = logits.argsort(dim=1, descending=True)
+    hit1 = (order[:, 0] == labels).float().mean() * 100
+    hit5 = (order[:, :5] == labels[:, None]).any(dim=1).float().mean() * 100
+    assert torch.isclose(t1, hit1)
+    assert torch.isclose(t5, hit5)
+    # tiny class count: k clamps to num_classes
+    small = torch.randn(8, 3)
+    lab = torch.randint(3, (8,))
+    _, t5s = metrics.topk(small, lab, topk=(1, 5))
+    assert t5s == 100.0
