@@ -8,7 +8,10 @@ this chip).  ``tools/conv_microbench.py`` measures these against MIOpen's
 igemm solvers shape-by-shape; dispatch is controlled by:
 
 * env ``BYOL_MFMA_CONV1X1`` = "0" (off), "1" (always), "auto" (default:
-  on for the shapes the microbench showed at parity-or-better).
+  the measured per-shape tables below — fwd 8/12, dgrad 11/12, wgrad 3/12
+  ResNet-50 shapes as of round 2), or "autotune" (runtime per-shape
+  ours-vs-MIOpen selection, cudnn.benchmark-style; cache via
+  ``BYOL_AUTOTUNE_CACHE``).  ``--conv-dispatch`` maps to these.
 """
 
 import os
