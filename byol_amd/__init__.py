@@ -6,4 +6,4 @@ hand-written CDNA4 (gfx950) HIP kernels for the hot ops + RCCL over xGMI
 for data parallelism.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"  # round 2: measured conv/BN dispatch, bf16 BN, aug v2, HYBRID find

@@ -29,7 +29,7 @@ sources = [
 
 setup(
     name="byol_amd",
-    version="0.1.0",
+    version="0.2.0",
     packages=["byol_amd"],
     ext_modules=[
         CUDAExtension(
